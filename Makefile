@@ -1,0 +1,42 @@
+# cea_amd build — native shim + HIP kernels built IN-TREE so the .so files
+# travel to GPU nodes with the source snapshot.
+# Parity with the reference Makefile targets (test/format/vet/presubmit,
+# /root/reference/Makefile:15-35) adapted to the Python+C++ toolchain.
+
+ROCM ?= /opt/rocm
+HIPCC ?= $(ROCM)/bin/hipcc
+CXX ?= g++
+
+SMI_SO := cea_amd/amdsmi/libceaamd_smi.so
+GPU_SO := cea_amd/ops/libceaamd_gpu.so
+
+.PHONY: all smi gpu test gputest lint presubmit clean
+
+all: smi gpu
+
+smi: $(SMI_SO)
+
+$(SMI_SO): csrc/amdsmi_shim.cpp
+	$(CXX) -O2 -std=c++17 -Wall -shared -fPIC -pthread \
+	  -I$(ROCM)/include $< -o $@ \
+	  -L$(ROCM)/lib -lamd_smi -Wl,-rpath,$(ROCM)/lib
+
+gpu: $(GPU_SO)
+
+$(GPU_SO): csrc/gpu_ops.hip
+	$(HIPCC) --offload-arch=gfx950 -O3 -std=c++17 -Wall -shared -fPIC $< -o $@
+
+test:
+	python3 -m pytest tests/ -x -q -m "not gpu"
+
+gputest:
+	python3 -m pytest tests/ -x -q -m gpu
+
+lint:
+	python3 -m compileall -q cea_amd cmd tests bench.py __graft_entry__.py
+	python3 build_tools/boilerplate.py
+
+presubmit: lint all test
+
+clean:
+	rm -f $(SMI_SO) $(GPU_SO)

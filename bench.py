@@ -1,0 +1,227 @@
+#!/usr/bin/env python3
+"""cea_amd flagship benchmark: RCCL all_reduce bus bandwidth over xGMI.
+
+BASELINE.json metric: "rccl-tests all_reduce bus-bw (GB/s) at 1/2/4/8 GPUs;
+Allocate() p50 latency".  The reference ships nccl-tests harnesses with the
+sweep 1 MiB -> 512 MiB, factor 2, 5 warmup, 100 iters, no data check
+(/root/reference/gpudirect-tcpx/nccl-config.yaml:17,55-58); this bench runs
+the same protocol through torch.distributed (backend "nccl" IS RCCL on
+ROCm), one process per GPU over xGMI.
+
+Step semantics (the driver's timed contract): one step = one out-of-place
+all_reduce of the headline 512 MiB fp32 buffer (copy into the recv buffer +
+in-place collective — the same work rccl-tests' out-of-place mode times).
+The full message sweep runs outside the timed region and is reported in
+config.sweep for the scaling curve.
+
+bus-bw convention (nccl-tests): busbw = algbw * 2*(n-1)/n for all_reduce.
+For n==1 that factor degenerates to 0, so the single-GPU value reported is
+the out-of-place algbw (the d2d copy that a world-1 out-of-place all_reduce
+performs); scaling judgments should use the n>=2 points.
+
+Allocate() p50: measured in-process against the real device plugin gRPC
+service over a unix socket with the mock amdsmi backend (pure control-plane
+latency, no GPU needed), reported in config.allocate_p50_us.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--max-bytes", type=int, default=512 * 1024 * 1024)
+    p.add_argument("--min-bytes", type=int, default=1024 * 1024)
+    p.add_argument("--sweep-iters", type=int, default=100)
+    p.add_argument("--sweep-warmup", type=int, default=5)
+    p.add_argument("--no-sweep", action="store_true",
+                   help="skip the per-size sweep (headline size only)")
+    p.add_argument("--backend", default="", help="nccl|gloo (default: auto)")
+    return p.parse_args()
+
+
+def allocate_p50_us(iters: int = 300) -> float:
+    """p50 latency of the device-plugin Allocate RPC over a real unix-socket
+    gRPC round-trip (the latency-critical kubelet path, beta_plugin.go:56)."""
+    import tempfile
+    import threading
+
+    import cea_amd.amdsmi as amdsmi
+    from cea_amd.amdsmi.mock import MockAmdSmi, make_fake_dev
+    from cea_amd.deviceplugin.manager import AmdGPUManager, GPUConfig
+
+    import grpc
+    from cea_amd.kube import protos as api
+
+    with tempfile.TemporaryDirectory() as tmp:
+        dev = os.path.join(tmp, "dev")
+        make_fake_dev(dev, 8)
+        mock = MockAmdSmi(dev)
+        mock.init()
+        amdsmi.set_ops(mock)
+        plugin_dir = os.path.join(tmp, "plugin")
+        os.makedirs(plugin_dir)
+        mgr = AmdGPUManager(GPUConfig(), dev_directory=dev,
+                            plugin_directory=plugin_dir)
+        mgr.config.add_defaults_and_validate()
+        mgr.start()
+        t = threading.Thread(target=mgr.serve, daemon=True)
+        t.start()
+        sock = os.path.join(plugin_dir, mgr.socket_name)
+        deadline = time.time() + 5
+        while not os.path.exists(sock) and time.time() < deadline:
+            time.sleep(0.01)
+        channel = grpc.insecure_channel(f"unix://{sock}")
+        allocate = channel.unary_unary(
+            api.DP_ALLOCATE,
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=api.AllocateResponse.FromString,
+        )
+        req = api.AllocateRequest()
+        req.container_requests.add(devices_ids=["amdgpu0"])
+        lat = []
+        for _ in range(iters):
+            t0 = time.perf_counter()
+            allocate(req, timeout=5)
+            lat.append((time.perf_counter() - t0) * 1e6)
+        channel.close()
+        mgr.stop()
+        amdsmi.ops = None
+        return statistics.median(lat)
+
+
+def main():
+    args = parse_args()
+    import torch
+    import torch.distributed as dist
+
+    env_world = int(os.environ.get("WORLD_SIZE", "1"))
+    world = max(env_world, 1)
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    has_gpu = torch.cuda.is_available()
+    backend = args.backend or ("nccl" if has_gpu else "gloo")
+    device = torch.device("cpu")
+    if has_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+    dist.init_process_group(backend=backend, world_size=world, rank=rank)
+
+    n = world
+    elems = args.max_bytes // 4
+    send = torch.empty(elems, dtype=torch.float32, device=device)
+    send.uniform_()
+    recv = torch.empty_like(send)
+
+    def step(buf_send, buf_recv):
+        buf_recv.copy_(buf_send)
+        dist.all_reduce(buf_recv)
+
+    def sync():
+        if has_gpu:
+            torch.cuda.synchronize()
+
+    def busbw_factor(nranks: int) -> float:
+        return 2.0 * (nranks - 1) / nranks if nranks > 1 else 1.0
+
+    # ---- per-size sweep (reference protocol: -w 5 --iters 100 -c 0) ------
+    sweep = []
+    if not args.no_sweep:
+        size = args.min_bytes
+        while size <= args.max_bytes:
+            e = size // 4
+            s_send, s_recv = send[:e], recv[:e]
+            for _ in range(args.sweep_warmup):
+                step(s_send, s_recv)
+            sync()
+            dist.barrier()
+            t0 = time.perf_counter()
+            for _ in range(args.sweep_iters):
+                step(s_send, s_recv)
+            sync()
+            el = time.perf_counter() - t0
+            dist.barrier()
+            t_max = torch.tensor([el], dtype=torch.float64)
+            dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
+            t_iter = t_max.item() / args.sweep_iters
+            algbw = size / t_iter / 1e9
+            sweep.append(
+                {
+                    "bytes": size,
+                    "time_us": round(t_iter * 1e6, 2),
+                    "algbw_GBps": round(algbw, 2),
+                    "busbw_GBps": round(algbw * busbw_factor(n), 2),
+                }
+            )
+            size *= 2
+
+    # ---- timed region: K steps at the headline size ----------------------
+    for _ in range(args.warmup):
+        step(send, recv)
+    sync()
+    dist.barrier()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step(send, recv)
+    sync()
+    elapsed = time.perf_counter() - t0
+    dist.barrier()
+    t_max = torch.tensor([elapsed], dtype=torch.float64)
+    dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
+    elapsed = t_max.item()
+
+    ms_per_step = elapsed / args.steps * 1e3
+    algbw = args.max_bytes / (elapsed / args.steps) / 1e9
+    busbw = algbw * busbw_factor(n)
+
+    if rank == 0:
+        try:
+            p50 = round(allocate_p50_us(), 1)
+        except Exception as e:  # noqa: BLE001 - p50 is auxiliary
+            p50 = None
+        result = {
+            "metric": "rccl-tests all_reduce bus-bw (GB/s)",
+            "value": round(busbw, 2),
+            "unit": "GB/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "all_reduce_perf",
+                "collective": "all_reduce",
+                "message_bytes": args.max_bytes,
+                "mode": "out-of-place",
+                "global_batch": None,
+                "seq_len": None,
+                "parallelism": f"dp{n}",
+                "backend": backend,
+                "transport": "RCCL/xGMI" if has_gpu else "gloo(cpu-test)",
+                "busbw_factor": round(busbw_factor(n), 4),
+                "algbw_GBps": round(algbw, 2),
+                "sweep": sweep,
+                "allocate_p50_us": p50,
+            },
+        }
+        print(json.dumps(result))
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
