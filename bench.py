@@ -115,7 +115,8 @@ def main():
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29571")
-    dist.init_process_group(backend=backend, world_size=world, rank=rank)
+    kwargs = {"device_id": device} if has_gpu else {}
+    dist.init_process_group(backend=backend, world_size=world, rank=rank, **kwargs)
 
     n = world
     elems = args.max_bytes // 4
@@ -151,7 +152,7 @@ def main():
             sync()
             el = time.perf_counter() - t0
             dist.barrier()
-            t_max = torch.tensor([el], dtype=torch.float64)
+            t_max = torch.tensor([el], dtype=torch.float64, device=device)
             dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
             t_iter = t_max.item() / args.sweep_iters
             algbw = size / t_iter / 1e9
@@ -177,7 +178,7 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
     dist.barrier()
-    t_max = torch.tensor([elapsed], dtype=torch.float64)
+    t_max = torch.tensor([elapsed], dtype=torch.float64, device=device)
     dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
     elapsed = t_max.item()
 
