@@ -1,0 +1,186 @@
+"""Health-checker tests — parity with the reference's
+health_checker_test.go (TestCatchError:44, TestMonitorXidevent:359,
+TestUpdateLastHeartbeatTime:266, TestResetXIDConditionWithBackoff:297),
+driven by hand-built events + FakeKubeClient."""
+import json
+
+import pytest
+
+import cea_amd.amdsmi as amdsmi
+from cea_amd.amdsmi.iface import EVT_ECC_UNCORRECTABLE, EVT_THERMAL_THROTTLE, EVT_VMFAULT, Event
+from cea_amd.amdsmi.mock import MockAmdSmi, make_fake_dev
+from cea_amd.deviceplugin.health import CONDITION_TYPE, GPUHealthChecker
+from cea_amd.deviceplugin.manager import AmdGPUManager, GPUConfig
+from cea_amd.kube.client import FakeKubeClient, KubeError
+
+
+def make_env(tmp_path, num_gpus=2, partitions_per_gpu=1, compute_partition="SPX",
+             config=None, boot_id="boot-1"):
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, num_gpus * partitions_per_gpu)
+    mock = MockAmdSmi(dev, compute_partition=compute_partition,
+                      partitions_per_gpu=partitions_per_gpu)
+    mock.init()
+    amdsmi.set_ops(mock)
+    mgr = AmdGPUManager(config or GPUConfig(), dev_directory=dev,
+                        sysfs_root=str(tmp_path / "sys"))
+    mgr.config.add_defaults_and_validate()
+    mgr.start()
+    kube = FakeKubeClient(nodes=[{
+        "metadata": {"name": "node1", "labels": {}},
+        "status": {"conditions": []},
+    }])
+    boot_path = tmp_path / "boot_id"
+    boot_path.write_text(boot_id)
+    hc = GPUHealthChecker(mgr, kube, node_name="node1",
+                          boot_id_path=str(boot_path))
+    return mgr, mock, kube, hc
+
+
+def drain(q):
+    out = []
+    while not q.empty():
+        out.append(q.get_nowait())
+    return out
+
+
+def test_critical_event_with_uuid_marks_matching_device(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_ECC_UNCORRECTABLE))
+    unhealthy = drain(mgr.health)
+    assert [d.ID for d in unhealthy] == ["amdgpu1"]
+    assert unhealthy[0].health == "Unhealthy"
+    # node condition carries the code set + boot id
+    cond = [c for c in kube.nodes["node1"]["status"]["conditions"]
+            if c["type"] == CONDITION_TYPE][0]
+    assert json.loads(cond["reason"]) == [EVT_ECC_UNCORRECTABLE]
+    assert cond["message"] == "boot-1"
+    assert len(kube.events) == 1
+
+
+def test_critical_event_without_uuid_marks_all(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=3)
+    hc.catch_error(Event(device_uuid="", code=EVT_ECC_UNCORRECTABLE))
+    ids = sorted(d.ID for d in drain(mgr.health))
+    assert ids == ["amdgpu0", "amdgpu1", "amdgpu2"]
+
+
+def test_monitor_only_event_sets_condition_not_health(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.catch_error(Event(device_uuid="mock-uuid-0", code=EVT_VMFAULT))
+    assert drain(mgr.health) == []
+    cond = [c for c in kube.nodes["node1"]["status"]["conditions"]
+            if c["type"] == CONDITION_TYPE][0]
+    assert json.loads(cond["reason"]) == [EVT_VMFAULT]
+
+
+def test_non_critical_event_ignored(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.catch_error(Event(device_uuid="", code=EVT_THERMAL_THROTTLE))
+    assert drain(mgr.health) == []
+    assert kube.nodes["node1"]["status"]["conditions"] == []
+
+
+def test_condition_merges_codes(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.catch_error(Event(device_uuid="", code=EVT_VMFAULT))
+    hc.catch_error(Event(device_uuid="", code=EVT_ECC_UNCORRECTABLE))
+    cond = [c for c in kube.nodes["node1"]["status"]["conditions"]
+            if c["type"] == CONDITION_TYPE][0]
+    assert json.loads(cond["reason"]) == sorted([EVT_VMFAULT, EVT_ECC_UNCORRECTABLE])
+
+
+def test_cpx_die_event_marks_all_partitions(tmp_path):
+    cfg = GPUConfig(compute_partition="cpx")
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=1, partitions_per_gpu=8,
+                                   compute_partition="CPX", config=cfg)
+    hc.catch_error(Event(device_uuid="mock-uuid-0", code=EVT_ECC_UNCORRECTABLE))
+    ids = sorted(d.ID for d in drain(mgr.health))
+    assert ids == [f"amdgpu0/xcd{i}" for i in range(8)]
+
+
+def test_reset_condition_on_boot_id_change(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path, boot_id="boot-2")
+    kube.nodes["node1"]["status"]["conditions"] = [{
+        "type": CONDITION_TYPE, "status": "True",
+        "reason": "[48]", "message": "boot-1",  # stale: previous boot
+    }]
+    assert hc.try_reset_condition()
+    assert kube.nodes["node1"]["status"]["conditions"] == []
+
+
+def test_reset_keeps_condition_same_boot(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path, boot_id="boot-1")
+    kube.nodes["node1"]["status"]["conditions"] = [{
+        "type": CONDITION_TYPE, "status": "True",
+        "reason": "[48]", "message": "boot-1",
+    }]
+    assert hc.try_reset_condition()
+    assert len(kube.nodes["node1"]["status"]["conditions"]) == 1
+
+
+def test_reset_retries_on_api_error(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path, boot_id="boot-2")
+    kube.nodes["node1"]["status"]["conditions"] = [{
+        "type": CONDITION_TYPE, "status": "True",
+        "reason": "[48]", "message": "boot-1",
+    }]
+    calls = {"n": 0}
+
+    def reactor(verb, resource, obj):
+        if verb == "get" and resource == "nodes":
+            calls["n"] += 1
+            if calls["n"] == 1:
+                raise KubeError(500, "boom")
+
+    kube.prepend_reactor(reactor)
+    assert not hc.try_reset_condition()
+    assert hc.try_reset_condition()
+    assert kube.nodes["node1"]["status"]["conditions"] == []
+
+
+def test_heartbeat_updates_timestamp(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    kube.nodes["node1"]["status"]["conditions"] = [{
+        "type": CONDITION_TYPE, "status": "True",
+        "reason": "[48]", "message": "boot-1",
+        "lastHeartbeatTime": "2000-01-01T00:00:00Z",
+    }]
+    hc.update_heartbeat()
+    cond = kube.nodes["node1"]["status"]["conditions"][0]
+    assert cond["lastHeartbeatTime"] != "2000-01-01T00:00:00Z"
+
+
+def test_ecc_poll_synthesizes_event(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.ecc_poll_interval_s = 0.05  # speed the 10 s poll up for the test
+    hc.start()
+    try:
+        mock.set_ecc_count(1, 5)
+        import time
+        deadline = time.time() + 3
+        found = []
+        while time.time() < deadline and not found:
+            found = drain(mgr.health)
+            time.sleep(0.05)
+        assert found, "ECC increase did not surface as a health event"
+        assert found[0].ID == "amdgpu1"
+    finally:
+        hc.stop()
+
+
+def test_event_loop_consumes_injected_events(tmp_path):
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.start()
+    try:
+        import time
+        mock.inject_event(Event(device_uuid="mock-uuid-0",
+                                code=EVT_ECC_UNCORRECTABLE))
+        deadline = time.time() + 3
+        found = []
+        while time.time() < deadline and not found:
+            found = drain(mgr.health)
+            time.sleep(0.05)
+        assert [d.ID for d in found] == ["amdgpu0"]
+    finally:
+        hc.stop()
