@@ -1,0 +1,44 @@
+"""Driver-version visibility: publish amdgpu/ROCm versions as node
+annotations.
+
+Parity: /root/reference/pkg/gpu/nvidia/version_visibility/version_visibility.go
+(:30-46, :67-86) — nvml.SystemGetDriverVersion ->
+cloud.google.com/cuda.driver-version.* annotations via server-side apply.
+AMD mapping: amdsmi driver info (amdgpu KMD) + ROCm release ->
+amd.com/rocm.driver-version.* annotations, same field-manager pattern.
+"""
+from __future__ import annotations
+
+import logging
+import re
+from typing import Dict
+
+from .. import amdsmi
+
+log = logging.getLogger(__name__)
+
+ANNOTATION_PREFIX = "amd.com/rocm.driver-version"
+FIELD_MANAGER = "amd-gpu-device-plugin"   # parity: version_visibility.go:67
+
+
+def parse_version(version: str) -> Dict[str, str]:
+    """'6.10.5' -> {major: 6, minor: 10, revision: 5, full: 6.10.5}
+    (parity: the major/minor/revision/full split, version_visibility.go:30-35).
+    Tolerates suffixes like '6.10.5-2009582.22.04'."""
+    out = {f"{ANNOTATION_PREFIX}.full": version}
+    m = re.match(r"^(\d+)\.(\d+)(?:\.(\d+))?", version)
+    if m:
+        out[f"{ANNOTATION_PREFIX}.major"] = m.group(1)
+        out[f"{ANNOTATION_PREFIX}.minor"] = m.group(2)
+        out[f"{ANNOTATION_PREFIX}.revision"] = m.group(3) or "0"
+    return out
+
+
+def publish_driver_version_annotations(kube_client, node_name: str) -> None:
+    """Parity: PublishDriverVersionAnnotations (version_visibility.go:38-46)."""
+    ops = amdsmi.get_ops()
+    version = ops.driver_version()
+    annotations = parse_version(version)
+    annotations["amd.com/rocm.release"] = ops.rocm_version()
+    kube_client.apply_node_annotations(node_name, annotations, FIELD_MANAGER)
+    log.info("published driver version annotations: %s", annotations)
