@@ -1,0 +1,172 @@
+"""partition_gpu — one-shot node job that drives MI355X compute/memory
+partition modes.
+
+Role parity: /root/reference/partition_gpu/partition_gpu.go (467 LoC), which
+drives `nvidia-smi mig` (enable MIG + reboot on Ampere, destroy/create
+GI/CI profiles, idempotency via parsing `nvidia-smi mig -lgi`).  The MI355X
+mechanism is simpler and runtime-switchable: `amd-smi set --gpu all
+--compute-partition {SPX|DPX|CPX}` (+ `--memory-partition NPS1/NPS2`), no
+reboot, but the set can fail with "busy" while KFD processes hold the GPU —
+handled with a bounded retry instead of the reference's SIGRTMIN+5 reboot
+(partition_gpu.go:297-300).
+
+Idempotency: parse `amd-smi static --gpu all --partition` (the analog of the
+parseLGIOutput table state machine, partition_gpu.go:394-444) and exit 0 if
+every GPU already shows the desired accelerator + memory partition.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import re
+import subprocess
+import time
+from typing import Callable, Dict, List, Optional, Tuple
+
+log = logging.getLogger(__name__)
+
+DEFAULT_CONFIG_PATH = "/etc/amd/gpu_config.json"
+
+VALID_COMPUTE = ("SPX", "DPX", "CPX")
+VALID_MEMORY = ("NPS1", "NPS2", "NPS4")
+
+# partitions per die by mode — parity with partitionSizeMaxCount
+# (partition_gpu.go:91-139)
+PARTITION_COUNT = {"SPX": 1, "DPX": 2, "CPX": 8}
+
+BUSY_RETRIES = 12
+BUSY_RETRY_DELAY_S = 10.0
+
+Runner = Callable[[List[str]], Tuple[int, str]]
+
+
+def default_runner(cmd: List[str]) -> Tuple[int, str]:
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=300)
+    return r.returncode, r.stdout + r.stderr
+
+
+def parse_partition_config(path: str) -> Tuple[str, str]:
+    """'cpx-nps1' (or ComputePartition/MemoryPartition keys) -> ('CPX','NPS1').
+    Parity: the config read at partition_gpu.go:157-178."""
+    with open(path) as f:
+        raw = json.load(f)
+    spec = raw.get("ComputePartition", raw.get("GPUPartitionSize", "")) or "spx"
+    parts = spec.strip().upper().split("-")
+    compute = parts[0]
+    memory = parts[1] if len(parts) > 1 else raw.get("MemoryPartition", "NPS1").upper()
+    if compute not in VALID_COMPUTE:
+        raise ValueError(f"invalid compute partition {compute!r}: want {VALID_COMPUTE}")
+    if memory not in VALID_MEMORY:
+        raise ValueError(f"invalid memory partition {memory!r}: want {VALID_MEMORY}")
+    return compute, memory
+
+
+GPU_RE = re.compile(r"^GPU:\s*(\d+)\s*$")
+KV_RE = re.compile(r"^\s+([A-Z_]+):\s*(\S+)\s*$")
+
+
+def parse_partition_status(text: str) -> List[Dict[str, str]]:
+    """Parse `amd-smi static --gpu all --partition` human output into
+    [{gpu, accelerator_partition, memory_partition, partition_id}].
+    The regex state machine is the analog of parseLGIOutput
+    (partition_gpu.go:394-444)."""
+    out: List[Dict[str, str]] = []
+    cur: Optional[Dict[str, str]] = None
+    for line in text.splitlines():
+        m = GPU_RE.match(line.strip()) or GPU_RE.match(line)
+        if line.strip().startswith("GPU:"):
+            gpu = line.split(":", 1)[1].strip()
+            cur = {"gpu": gpu}
+            out.append(cur)
+            continue
+        m = KV_RE.match(line)
+        if m and cur is not None:
+            key = m.group(1).lower()
+            if key in ("accelerator_partition", "compute_partition",
+                       "memory_partition", "partition_id"):
+                cur[key] = m.group(2)
+    # normalize: older amd-smi prints COMPUTE_PARTITION
+    for d in out:
+        if "accelerator_partition" not in d and "compute_partition" in d:
+            d["accelerator_partition"] = d["compute_partition"]
+    return [d for d in out if "accelerator_partition" in d]
+
+
+def check_desired(states: List[Dict[str, str]], compute: str, memory: str) -> bool:
+    """Uniformity + desired-mode check (parity: checkDesired,
+    partition_gpu.go:446-458)."""
+    if not states:
+        return False
+    for s in states:
+        if s.get("accelerator_partition", "").upper() != compute:
+            return False
+        mp = s.get("memory_partition", "").upper()
+        if mp and mp != memory:
+            return False
+    return True
+
+
+class PartitionError(RuntimeError):
+    pass
+
+
+def current_partition_status(runner: Runner) -> List[Dict[str, str]]:
+    rc, out = runner(["amd-smi", "static", "--gpu", "all", "--partition"])
+    if rc != 0:
+        raise PartitionError(f"amd-smi static failed (rc={rc}): {out[:500]}")
+    return parse_partition_status(out)
+
+
+def _set_with_busy_retry(runner: Runner, cmd: List[str]) -> None:
+    """amd-smi set fails while KFD processes hold the GPU; bounded retry
+    (the drain-and-retry replacing the reference's node reboot)."""
+    for attempt in range(BUSY_RETRIES):
+        rc, out = runner(cmd)
+        if rc == 0:
+            return
+        lowered = out.lower()
+        if "busy" in lowered or "in use" in lowered:
+            log.warning("GPU busy (attempt %d/%d): waiting for workloads to "
+                        "drain", attempt + 1, BUSY_RETRIES)
+            time.sleep(BUSY_RETRY_DELAY_S)
+            continue
+        raise PartitionError(f"{' '.join(cmd)} failed (rc={rc}): {out[:500]}")
+    raise PartitionError(f"{' '.join(cmd)}: GPU still busy after "
+                         f"{BUSY_RETRIES} attempts")
+
+
+def run(config_path: str = DEFAULT_CONFIG_PATH,
+        runner: Runner = default_runner) -> bool:
+    """Main flow (parity: main, partition_gpu.go:157-236).  Returns True if a
+    mode change was applied, False if already in the desired state."""
+    compute, memory = parse_partition_config(config_path)
+    states = current_partition_status(runner)
+    if not states:
+        raise PartitionError("amd-smi reported no GPUs")
+    log.info("desired: %s/%s; current: %s", compute, memory,
+             [(s.get("accelerator_partition"), s.get("memory_partition"))
+              for s in states])
+    if check_desired(states, compute, memory):
+        log.info("GPUs already partitioned as %s/%s; nothing to do",
+                 compute, memory)
+        return False
+
+    # memory partition first (it implies a KFD re-enumeration), then compute
+    needs_mem = any(
+        s.get("memory_partition", "").upper() not in ("", memory) for s in states
+    )
+    if needs_mem:
+        _set_with_busy_retry(
+            runner, ["amd-smi", "set", "--gpu", "all",
+                     "--memory-partition", memory])
+    _set_with_busy_retry(
+        runner, ["amd-smi", "set", "--gpu", "all",
+                 "--compute-partition", compute])
+
+    states = current_partition_status(runner)
+    if not check_desired(states, compute, memory):
+        raise PartitionError(
+            f"partitioning applied but verification failed: {states}")
+    log.info("partitioned all GPUs to %s/%s (%d logical devices/die)",
+             compute, memory, PARTITION_COUNT[compute])
+    return True
