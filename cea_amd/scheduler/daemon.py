@@ -1,0 +1,260 @@
+"""Gang scheduler daemon: scheduling-gate driven, topology-aware.
+
+Role parity: /root/reference/gke-topology-scheduler/schedule-daemon.py
+(814 LoC).  Protocol kept identical so workloads are drop-in:
+  * pods opt in with a scheduling gate named `gke.io/topology-aware-auto-*`
+    (schedule-daemon.py:760) — the daemon finds Pending gated pods, groups
+    them into jobs, and for each complete gang picks nodes minimizing
+    topology distance, then binds by REMOVING the gate and pinning the pod
+    with required nodeAffinity on kubernetes.io/hostname
+    (schedule-daemon.py:447-497);
+  * job grouping by job-name / jobset labels with owner-UID fallback
+    (:594-647); pods sorted by completion index (:119-150);
+  * node feasibility = Ready + taint-toleration compatible + free
+    cpu/memory/amd.com/gpu computed from running pods (:245-332).
+Resource key is amd.com/gpu (the reference hardcodes nvidia.com/gpu at
+:221,311,389).
+"""
+from __future__ import annotations
+
+import logging
+import re
+import time
+from collections import defaultdict
+from typing import Dict, List, Optional
+
+from ..deviceplugin import RESOURCE_NAME
+from .topology import CandidateNode, assign_pods, topo_key_from_labels
+
+log = logging.getLogger(__name__)
+
+GATE_PREFIX = "gke.io/topology-aware-auto-"   # parity schedule-daemon.py:760
+COMPLETION_INDEX_LABEL = "batch.kubernetes.io/job-completion-index"
+JOB_NAME_LABELS = (
+    "job-name",
+    "jobset.sigs.k8s.io/jobset-name",
+    "batch.kubernetes.io/job-name",
+)
+
+_QUANTITY_RE = re.compile(r"^(\d+(?:\.\d+)?)([a-zA-Z]*)$")
+_SUFFIX = {
+    "": 1, "k": 10**3, "M": 10**6, "G": 10**9, "T": 10**12, "P": 10**15,
+    "Ki": 2**10, "Mi": 2**20, "Gi": 2**30, "Ti": 2**40, "Pi": 2**50,
+}
+
+
+def parse_quantity(q) -> float:
+    """k8s resource quantity -> float (cpu in cores, memory in bytes)."""
+    if isinstance(q, (int, float)):
+        return float(q)
+    q = str(q)
+    if q.endswith("m"):
+        return float(q[:-1]) / 1000.0
+    m = _QUANTITY_RE.match(q)
+    if not m:
+        return 0.0
+    return float(m.group(1)) * _SUFFIX.get(m.group(2), 1)
+
+
+def pod_requests(pod: dict) -> Dict[str, float]:
+    """Summed container requests (parity :356-417)."""
+    out: Dict[str, float] = defaultdict(float)
+    for c in pod.get("spec", {}).get("containers", []):
+        reqs = c.get("resources", {}).get("requests", {}) or {}
+        for key, val in reqs.items():
+            out[key] += parse_quantity(val)
+    return dict(out)
+
+
+def has_topology_gate(pod: dict) -> Optional[str]:
+    for gate in pod.get("spec", {}).get("schedulingGates", []) or []:
+        name = gate.get("name", "")
+        if name.startswith(GATE_PREFIX):
+            return name
+    return None
+
+
+def job_key(pod: dict) -> str:
+    """Group key (parity :594-647): job labels, then owner UID, then the
+    pod's own name."""
+    meta = pod.get("metadata", {})
+    labels = meta.get("labels", {}) or {}
+    ns = meta.get("namespace", "default")
+    for lab in JOB_NAME_LABELS:
+        if lab in labels:
+            return f"{ns}/{labels[lab]}"
+    for owner in meta.get("ownerReferences", []) or []:
+        if owner.get("uid"):
+            return f"{ns}/{owner['uid']}"
+    return f"{ns}/{meta.get('name', '')}"
+
+
+def completion_index(pod: dict) -> int:
+    """Parity :119-150 — job completion index label, else trailing ordinal
+    in the pod name."""
+    labels = pod.get("metadata", {}).get("labels", {}) or {}
+    if COMPLETION_INDEX_LABEL in labels:
+        try:
+            return int(labels[COMPLETION_INDEX_LABEL])
+        except ValueError:
+            pass
+    m = re.search(r"(\d+)$", pod.get("metadata", {}).get("name", ""))
+    return int(m.group(1)) if m else 0
+
+
+def node_is_ready(node: dict) -> bool:
+    for c in node.get("status", {}).get("conditions", []) or []:
+        if c.get("type") == "Ready":
+            return c.get("status") == "True"
+    return False
+
+
+def tolerates(pod: dict, node: dict) -> bool:
+    """NoSchedule/NoExecute taint compatibility (parity :245-303)."""
+    taints = node.get("spec", {}).get("taints", []) or []
+    tolerations = pod.get("spec", {}).get("tolerations", []) or []
+    for taint in taints:
+        if taint.get("effect") not in ("NoSchedule", "NoExecute"):
+            continue
+        ok = False
+        for tol in tolerations:
+            op = tol.get("operator", "Equal")
+            if tol.get("key") in (None, "", taint.get("key")) and op == "Exists":
+                ok = True
+            elif (tol.get("key") == taint.get("key")
+                  and op == "Equal"
+                  and tol.get("value", "") == taint.get("value", "")):
+                ok = True
+            if ok and tol.get("effect") in (None, "", taint.get("effect")):
+                break
+            ok = False
+        if not ok:
+            return False
+    return True
+
+
+def node_free_resources(node: dict, pods_on_node: List[dict]) -> Dict[str, float]:
+    """allocatable - sum(requests of non-terminal pods) (parity :305-332)."""
+    free: Dict[str, float] = {
+        k: parse_quantity(v)
+        for k, v in (node.get("status", {}).get("allocatable", {}) or {}).items()
+    }
+    for pod in pods_on_node:
+        phase = pod.get("status", {}).get("phase", "")
+        if phase in ("Succeeded", "Failed"):
+            continue
+        for k, v in pod_requests(pod).items():
+            if k in free:
+                free[k] -= v
+    return free
+
+
+def pods_fit_count(free: Dict[str, float], req: Dict[str, float]) -> int:
+    """How many pods with request `req` fit into `free`."""
+    count = float("inf")
+    for k, v in req.items():
+        if v <= 0:
+            continue
+        count = min(count, free.get(k, 0.0) // v)
+    return int(count) if count != float("inf") else 2**30
+
+
+def bind_pod(kube, pod: dict, node_name: str) -> None:
+    """Drop the topology gate + inject required nodeAffinity on
+    kubernetes.io/hostname (parity :447-497)."""
+    ns = pod["metadata"].get("namespace", "default")
+    name = pod["metadata"]["name"]
+    spec = pod.setdefault("spec", {})
+    spec["schedulingGates"] = [
+        g for g in spec.get("schedulingGates", []) or []
+        if not g.get("name", "").startswith(GATE_PREFIX)
+    ]
+    affinity = spec.setdefault("affinity", {})
+    node_affinity = affinity.setdefault("nodeAffinity", {})
+    required = node_affinity.setdefault(
+        "requiredDuringSchedulingIgnoredDuringExecution", {"nodeSelectorTerms": []}
+    )
+    required["nodeSelectorTerms"] = [{
+        "matchExpressions": [{
+            "key": "kubernetes.io/hostname",
+            "operator": "In",
+            "values": [node_name],
+        }]
+    }]
+    kube.replace_pod(ns, name, pod)
+    log.info("bound %s/%s -> %s", ns, name, node_name)
+
+
+class TopologyScheduler:
+    def __init__(self, kube, interval_s: float = 5.0,
+                 gate_cooloff_s: float = 60.0):
+        self.kube = kube
+        self.interval_s = interval_s
+        self.gate_cooloff_s = gate_cooloff_s   # parity :777-807
+        self._last_attempt: Dict[str, float] = {}
+        self._stop = False
+
+    def schedule_once(self) -> int:
+        """One pass; returns the number of pods bound."""
+        pods = self.kube.list_pods(field_selector="status.phase=Pending")
+        gated = [p for p in pods if has_topology_gate(p)]
+        if not gated:
+            return 0
+        jobs: Dict[str, List[dict]] = defaultdict(list)
+        for p in gated:
+            jobs[job_key(p)].append(p)
+
+        nodes = [n for n in self.kube.list_nodes() if node_is_ready(n)]
+        all_pods = self.kube.list_pods()
+        pods_by_node: Dict[str, List[dict]] = defaultdict(list)
+        for p in all_pods:
+            node_name = p.get("spec", {}).get("nodeName")
+            if node_name:
+                pods_by_node[node_name].append(p)
+
+        bound = 0
+        now = time.monotonic()
+        for key, job_pods in jobs.items():
+            last = self._last_attempt.get(key, 0)
+            if now - last < self.gate_cooloff_s and last > 0:
+                continue
+            self._last_attempt[key] = now
+            job_pods.sort(key=completion_index)
+            req = pod_requests(job_pods[0])
+            candidates = []
+            for n in nodes:
+                if not tolerates(job_pods[0], n):
+                    continue
+                free = node_free_resources(n, pods_by_node[n["metadata"]["name"]])
+                cap = pods_fit_count(free, req)
+                if cap > 0:
+                    candidates.append(CandidateNode(
+                        name=n["metadata"]["name"],
+                        topo=topo_key_from_labels(
+                            n["metadata"].get("labels", {}) or {}),
+                        capacity=cap,
+                    ))
+            assignment = assign_pods(len(job_pods), candidates)
+            if assignment is None:
+                log.info("job %s: %d pods do not fit on %d candidate nodes; "
+                         "waiting", key, len(job_pods), len(candidates))
+                continue
+            for pod, node_name in zip(job_pods, assignment):
+                try:
+                    bind_pod(self.kube, pod, node_name)
+                    bound += 1
+                    pods_by_node[node_name].append(pod)
+                except Exception as e:  # noqa: BLE001
+                    log.error("bind failed for %s: %s",
+                              pod["metadata"].get("name"), e)
+        return bound
+
+    def run_forever(self) -> None:
+        log.info("topology scheduler loop starting (interval %.0fs)",
+                 self.interval_s)
+        while not self._stop:
+            try:
+                self.schedule_once()
+            except Exception as e:  # noqa: BLE001
+                log.error("scheduling pass failed: %s", e)
+            time.sleep(self.interval_s)

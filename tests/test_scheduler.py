@@ -1,0 +1,203 @@
+"""Topology scheduler tests: distance model, contiguous-window assignment,
+gang scheduling end-to-end against FakeKubeClient."""
+import pytest
+
+from cea_amd.kube.client import FakeKubeClient
+from cea_amd.scheduler import daemon as d
+from cea_amd.scheduler import topology as topo
+from cea_amd.scheduler.labeler import label_node, parse_physical_host, topology_from_env
+from cea_amd.scheduler.topology import CandidateNode, TopoKey, assign_pods
+
+
+def K(b, s="", h=""):
+    return TopoKey(b, s, h)
+
+
+def test_distance_weights():
+    assert topo.distance(K("b1", "s1", "h1"), K("b1", "s1", "h1")) == 0
+    assert topo.distance(K("b1", "s1", "h1"), K("b1", "s1", "h2")) == 100
+    assert topo.distance(K("b1", "s1", "h1"), K("b1", "s2", "h2")) == 10100
+    assert topo.distance(K("b1", "s1", "h1"), K("b2", "s2", "h2")) == 1010100
+
+
+def test_assign_prefers_same_host_block():
+    nodes = [
+        CandidateNode("far", K("b2", "s1", "h9"), 1),
+        CandidateNode("n1", K("b1", "s1", "h1"), 1),
+        CandidateNode("n2", K("b1", "s1", "h1"), 1),
+        CandidateNode("n3", K("b1", "s2", "h2"), 1),
+    ]
+    out = assign_pods(2, nodes)
+    assert sorted(out) == ["n1", "n2"]
+    # 3 pods: must include n3 (same block) over `far` (other block)
+    out = assign_pods(3, nodes)
+    assert sorted(out) == ["n1", "n2", "n3"]
+
+
+def test_assign_packs_one_node_when_possible():
+    # a single node with spare capacity beats spanning: distance 0
+    nodes = [
+        CandidateNode("big", K("b2", "s1", "h9"), 8),
+        CandidateNode("n1", K("b1", "s1", "h1"), 1),
+        CandidateNode("n2", K("b1", "s1", "h2"), 1),
+    ]
+    assert assign_pods(3, nodes) == ["big"] * 3
+
+
+def test_assign_uses_capacity_and_fails_when_too_big():
+    nodes = [CandidateNode("n1", K("b1"), 4)]
+    assert assign_pods(4, nodes) == ["n1"] * 4
+    assert assign_pods(5, nodes) is None
+    assert assign_pods(0, nodes) is None
+
+
+def test_quantity_parsing():
+    assert d.parse_quantity("500m") == 0.5
+    assert d.parse_quantity("2") == 2
+    assert d.parse_quantity("1Gi") == 2**30
+    assert d.parse_quantity("8") == 8
+    assert d.parse_quantity(4) == 4
+
+
+def make_node(name, labels=None, gpus=8, cpu="64", mem="512Gi", taints=None):
+    return {
+        "metadata": {"name": name, "labels": labels or {}},
+        "spec": {"taints": taints or []},
+        "status": {
+            "conditions": [{"type": "Ready", "status": "True"}],
+            "allocatable": {"cpu": cpu, "memory": mem, "amd.com/gpu": str(gpus)},
+        },
+    }
+
+
+def make_pod(name, ns="default", job="j1", idx=None, gpus=8, gated=True,
+             node=None, phase="Pending"):
+    labels = {"job-name": job}
+    if idx is not None:
+        labels["batch.kubernetes.io/job-completion-index"] = str(idx)
+    pod = {
+        "metadata": {"name": name, "namespace": ns, "labels": labels},
+        "spec": {
+            "containers": [{
+                "name": "main",
+                "resources": {"requests": {"amd.com/gpu": str(gpus),
+                                           "cpu": "4", "memory": "16Gi"}},
+            }],
+            "schedulingGates": (
+                [{"name": "gke.io/topology-aware-auto-j1"}] if gated else []
+            ),
+        },
+        "status": {"phase": phase},
+    }
+    if node:
+        pod["spec"]["nodeName"] = node
+    return pod
+
+
+def topo_labels(b, s, h):
+    return {
+        "topology.cea-amd.io/block": b,
+        "topology.cea-amd.io/subblock": s,
+        "topology.cea-amd.io/host": h,
+    }
+
+
+def test_gang_schedules_to_closest_nodes():
+    kube = FakeKubeClient(
+        nodes=[
+            make_node("nodeA", topo_labels("b1", "s1", "h1")),
+            make_node("nodeB", topo_labels("b1", "s1", "h2")),
+            make_node("nodeC", topo_labels("b2", "s9", "h9")),
+        ],
+        pods=[make_pod("j1-0", idx=0), make_pod("j1-1", idx=1)],
+    )
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 2
+    p0 = kube.get_pod("default", "j1-0")
+    p1 = kube.get_pod("default", "j1-1")
+    assert p0["spec"]["schedulingGates"] == []
+    chosen = {
+        p["spec"]["affinity"]["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"
+        ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"][0]
+        for p in (p0, p1)
+    }
+    assert chosen == {"nodeA", "nodeB"}  # same subblock, not nodeC
+
+
+def test_gang_waits_when_not_fitting():
+    kube = FakeKubeClient(
+        nodes=[make_node("nodeA", topo_labels("b1", "s1", "h1"))],
+        pods=[make_pod("j1-0", idx=0), make_pod("j1-1", idx=1)],
+    )
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 0  # 2x8 GPUs don't fit on one 8-GPU node
+    assert d.has_topology_gate(kube.get_pod("default", "j1-0"))
+
+
+def test_running_pods_consume_capacity():
+    kube = FakeKubeClient(
+        nodes=[
+            make_node("nodeA", topo_labels("b1", "s1", "h1")),
+            make_node("nodeB", topo_labels("b1", "s1", "h2")),
+        ],
+        pods=[
+            make_pod("busy-0", job="other", gated=False, node="nodeA",
+                     phase="Running"),
+            make_pod("j1-0", idx=0),
+        ],
+    )
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 1
+    p = kube.get_pod("default", "j1-0")
+    vals = p["spec"]["affinity"]["nodeAffinity"][
+        "requiredDuringSchedulingIgnoredDuringExecution"
+    ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"]
+    assert vals == ["nodeB"]  # nodeA's GPUs are taken by busy-0
+
+
+def test_taints_respected():
+    kube = FakeKubeClient(
+        nodes=[
+            make_node("tainted", topo_labels("b1", "s1", "h1"),
+                      taints=[{"key": "dedicated", "value": "x",
+                               "effect": "NoSchedule"}]),
+            make_node("open", topo_labels("b1", "s1", "h2")),
+        ],
+        pods=[make_pod("j1-0", idx=0)],
+    )
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 1
+    p = kube.get_pod("default", "j1-0")
+    vals = p["spec"]["affinity"]["nodeAffinity"][
+        "requiredDuringSchedulingIgnoredDuringExecution"
+    ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"]
+    assert vals == ["open"]
+
+
+def test_completion_index_order():
+    assert d.completion_index(make_pod("x", idx=7)) == 7
+    assert d.completion_index(make_pod("worker-12")) == 12
+
+
+def test_job_grouping():
+    p1 = make_pod("a", job="jobX")
+    p2 = make_pod("b", job="jobX")
+    p3 = {"metadata": {"name": "c", "namespace": "default",
+                       "ownerReferences": [{"uid": "u1"}]}, "spec": {}}
+    assert d.job_key(p1) == d.job_key(p2)
+    assert d.job_key(p3) == "default/u1"
+
+
+def test_labeler():
+    assert parse_physical_host("/blk/sb/h") == {
+        "block": "blk", "subblock": "sb", "host": "h"}
+    assert parse_physical_host("") is None
+    assert topology_from_env({"TOPOLOGY_BLOCK": "b", "TOPOLOGY_SUBBLOCK": "s",
+                              "TOPOLOGY_HOST": "h"}) == {
+        "block": "b", "subblock": "s", "host": "h"}
+    kube = FakeKubeClient(nodes=[{"metadata": {"name": "n1", "labels": {}},
+                                  "status": {}}])
+    label_node(kube, "n1", {"block": "b", "subblock": "s", "host": "h"})
+    assert kube.nodes["n1"]["metadata"]["labels"][
+        "topology.cea-amd.io/block"] == "b"
