@@ -9,10 +9,11 @@ CXX ?= g++
 
 SMI_SO := cea_amd/amdsmi/libceaamd_smi.so
 GPU_SO := cea_amd/ops/libceaamd_gpu.so
+RCCL_BENCH := cea_amd/bin/all_reduce_perf
 
-.PHONY: all smi gpu test gputest lint presubmit clean
+.PHONY: all smi gpu rcclbench test gputest lint presubmit clean
 
-all: smi gpu
+all: smi gpu rcclbench
 
 smi: $(SMI_SO)
 
@@ -32,6 +33,13 @@ test:
 gputest:
 	python3 -m pytest tests/ -x -q -m gpu
 
+rcclbench: $(RCCL_BENCH)
+
+$(RCCL_BENCH): csrc/all_reduce_perf.cpp
+	mkdir -p cea_amd/bin
+	$(HIPCC) --offload-arch=gfx950 -O3 -std=c++17 -Wall $< -o $@ \
+	  -I$(ROCM)/include -L$(ROCM)/lib -lrccl -Wl,-rpath,$(ROCM)/lib
+
 lint:
 	python3 -m compileall -q cea_amd cmd tests bench.py __graft_entry__.py
 	python3 build_tools/boilerplate.py
@@ -39,4 +47,4 @@ lint:
 presubmit: lint all test
 
 clean:
-	rm -f $(SMI_SO) $(GPU_SO)
+	rm -f $(SMI_SO) $(GPU_SO) $(RCCL_BENCH)

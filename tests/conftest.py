@@ -1,3 +1,5 @@
+"""Shared pytest config: gpu marker registration + package path + global
+injection-point hygiene."""
 import os
 import sys
 
