@@ -1,0 +1,116 @@
+#!/bin/bash
+# amdgpu driver + ROCm userspace installer for Ubuntu nodes.
+#
+# Role parity: /root/reference/nvidia-driver-installer/ubuntu/entrypoint.sh
+# (cache check keyed on kernel+driver version :33-51, kernel headers :70-74,
+# installer run :122-135, verify :149-156, host ld.so.conf :158-163).
+# MI355X redesign: no proprietary .run installer and no overlayfs trick —
+# amdgpu-dkms builds against the node kernel from the ROCm apt repo, and the
+# ROCm userspace the device plugin mounts into pods (hip runtime, rocm-smi,
+# libamd_smi, librccl, rocBLAS/hipBLASLt/MIOpen) is copied into
+# ROOT_MOUNT_DIR/home/kubernetes/bin/amd.
+set -o errexit
+set -o pipefail
+set -u
+
+set -x
+
+ROCM_VERSION="${ROCM_VERSION:-7.2}"
+AMDGPU_DRIVER_VERSION="${AMDGPU_DRIVER_VERSION:-30.20}"
+ROOT_MOUNT_DIR="${ROOT_MOUNT_DIR:-/root}"
+INSTALL_DIR_HOST="${INSTALL_DIR_HOST:-/home/kubernetes/bin/amd}"
+INSTALL_DIR="${ROOT_MOUNT_DIR}${INSTALL_DIR_HOST}"
+CACHE_FILE="${INSTALL_DIR}/.cache"
+KERNEL_VERSION="$(uname -r)"
+
+check_cached_version() {
+  # Parity: cache keyed on (kernel, driver version), entrypoint.sh:33-51.
+  [[ -f "${CACHE_FILE}" ]] || return 1
+  grep -q "^CACHE_KERNEL_VERSION=${KERNEL_VERSION}$" "${CACHE_FILE}" || return 1
+  grep -q "^CACHE_AMDGPU_VERSION=${AMDGPU_DRIVER_VERSION}$" "${CACHE_FILE}" || return 1
+  grep -q "^CACHE_ROCM_VERSION=${ROCM_VERSION}$" "${CACHE_FILE}" || return 1
+  echo "amdgpu ${AMDGPU_DRIVER_VERSION} + ROCm ${ROCM_VERSION} already installed for ${KERNEL_VERSION}"
+  return 0
+}
+
+update_cached_version() {
+  cat >"${CACHE_FILE}" <<EOF
+CACHE_KERNEL_VERSION=${KERNEL_VERSION}
+CACHE_AMDGPU_VERSION=${AMDGPU_DRIVER_VERSION}
+CACHE_ROCM_VERSION=${ROCM_VERSION}
+EOF
+}
+
+install_kernel_headers() {
+  # Parity: entrypoint.sh:70-74.
+  apt-get update
+  apt-get install -y "linux-headers-${KERNEL_VERSION}" || \
+    apt-get install -y "linux-headers-generic"
+}
+
+setup_repos() {
+  local ub_codename
+  ub_codename="$(. /etc/os-release && echo "${VERSION_CODENAME}")"
+  mkdir -p /etc/apt/keyrings
+  curl -fsSL https://repo.radeon.com/rocm/rocm.gpg.key | \
+    gpg --dearmor -o /etc/apt/keyrings/rocm.gpg
+  echo "deb [arch=amd64 signed-by=/etc/apt/keyrings/rocm.gpg] \
+https://repo.radeon.com/amdgpu/${AMDGPU_DRIVER_VERSION}/ubuntu ${ub_codename} main" \
+    > /etc/apt/sources.list.d/amdgpu.list
+  echo "deb [arch=amd64 signed-by=/etc/apt/keyrings/rocm.gpg] \
+https://repo.radeon.com/rocm/apt/${ROCM_VERSION} ${ub_codename} main" \
+    > /etc/apt/sources.list.d/rocm.list
+  apt-get update
+}
+
+build_and_load_kmd() {
+  # dkms builds the amdgpu kernel module against the node kernel.
+  DEBIAN_FRONTEND=noninteractive apt-get install -y "amdgpu-dkms"
+  dkms autoinstall -k "${KERNEL_VERSION}"
+  modprobe amdgpu
+}
+
+install_rocm_userspace() {
+  DEBIAN_FRONTEND=noninteractive apt-get install -y \
+    rocm-core rocminfo rocm-smi-lib amd-smi-lib \
+    hip-runtime-amd rocblas hipblaslt miopen-hip rccl
+  mkdir -p "${INSTALL_DIR}/lib64" "${INSTALL_DIR}/bin" "${INSTALL_DIR}/.info"
+  # The subset GPU pods need, mounted read-only by the device plugin
+  # (manager.mounts(): /home/kubernetes/bin/amd -> /usr/local/amd).
+  cp -a /opt/rocm/lib/*.so* "${INSTALL_DIR}/lib64/" 2>/dev/null || true
+  for tool in rocminfo rocm-smi amd-smi hipconfig; do
+    [[ -x "/opt/rocm/bin/${tool}" ]] && cp -a "/opt/rocm/bin/${tool}" "${INSTALL_DIR}/bin/"
+  done
+  cp -a /opt/rocm/.info/version "${INSTALL_DIR}/.info/version" 2>/dev/null || true
+}
+
+verify_installation() {
+  # Parity: nvidia-smi verify + device-node check, entrypoint.sh:149-156.
+  [[ -e "${ROOT_MOUNT_DIR}/dev/kfd" ]] || { echo "/dev/kfd missing"; return 1; }
+  ls "${ROOT_MOUNT_DIR}"/dev/dri/renderD* >/dev/null
+  "${INSTALL_DIR}/bin/rocminfo" | grep -q "gfx950"
+  "${INSTALL_DIR}/bin/amd-smi" list
+}
+
+update_host_ld_cache() {
+  # Parity: entrypoint.sh:158-163.
+  echo "${INSTALL_DIR_HOST}/lib64" > "${ROOT_MOUNT_DIR}/etc/ld.so.conf.d/amd.conf"
+  ldconfig -r "${ROOT_MOUNT_DIR}"
+}
+
+main() {
+  if check_cached_version; then
+    verify_installation
+    exit 0
+  fi
+  install_kernel_headers
+  setup_repos
+  build_and_load_kmd
+  install_rocm_userspace
+  verify_installation
+  update_host_ld_cache
+  update_cached_version
+  echo "amdgpu + ROCm install complete"
+}
+
+main "$@"
