@@ -115,6 +115,9 @@ def main():
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29571")
+    # dmabuf IPC: required for RCCL cross-process CUDA-tensor sharing on
+    # hosts whose driver only supports dmabuf IPC
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     kwargs = {"device_id": device} if has_gpu else {}
     dist.init_process_group(backend=backend, world_size=world, rank=rank, **kwargs)
 
