@@ -310,7 +310,20 @@ class AmdGPUManager:
             self.plugin_directory, api.KUBELET_SOCKET
         )
         restarts = 0
+        first = True
         while not self._stop.is_set():
+            if not first:
+                # A restart trigger fired: re-run discovery so hot-added
+                # GPUs (or recreated partitions) are advertised
+                # (parity: the reference re-Starts the manager after its
+                # watchdog fires, manager.go:511-544).
+                try:
+                    self.start()
+                except Exception as e:  # noqa: BLE001
+                    log.error("restart discovery failed (retrying): %s", e)
+                    time.sleep(2)
+                    continue
+            first = False
             socket_path = os.path.join(self.plugin_directory, self.socket_name)
             try:
                 os.unlink(socket_path)
