@@ -127,9 +127,21 @@ def main():
     send.uniform_()
     recv = torch.empty_like(send)
 
-    def step(buf_send, buf_recv):
-        buf_recv.copy_(buf_send)
-        dist.all_reduce(buf_recv)
+    # On GPU the out-of-place copy uses the in-tree float4 HIP kernel
+    # (cea_amd/ops, gfx950) — the native path is mandatory on GPU nodes.
+    if has_gpu:
+        from cea_amd.ops import native
+
+        native.assert_native_available()
+
+        def step(buf_send, buf_recv):
+            native.copy_(buf_recv, buf_send)
+            dist.all_reduce(buf_recv)
+    else:
+
+        def step(buf_send, buf_recv):
+            buf_recv.copy_(buf_send)
+            dist.all_reduce(buf_recv)
 
     def sync():
         if has_gpu:

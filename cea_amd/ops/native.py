@@ -120,6 +120,31 @@ def vector_add(a, b, c, inject_fault: bool = False, stream=None) -> None:
     )
 
 
+def copy_(dst, src, stream=None) -> None:
+    """Async d2d copy dst <- src (contiguous CUDA tensors, same nbytes,
+    multiple of 16 bytes) with the float4-vectorized gfx950 kernel."""
+    import torch
+
+    lib = load_library()
+    assert dst.is_cuda and src.is_cuda
+    assert dst.is_contiguous() and src.is_contiguous()
+    nbytes = src.numel() * src.element_size()
+    assert dst.numel() * dst.element_size() == nbytes
+    sp = ctypes.c_void_p(
+        torch.cuda.current_stream().cuda_stream if stream is None else stream
+    )
+    _check(
+        lib,
+        lib.cea_copy(
+            ctypes.c_void_p(dst.data_ptr()),
+            ctypes.c_void_p(src.data_ptr()),
+            ctypes.c_long(nbytes),
+            sp,
+        ),
+        "cea_copy",
+    )
+
+
 def copy_bandwidth_gbps(bytes_n: int = 1 << 30, iters: int = 20) -> float:
     """Device d2d copy bandwidth in GB/s (read+write)."""
     import torch

@@ -83,6 +83,48 @@ def test_metrics_collector_on_hardware():
         amdsmi.ops = None
 
 
+def test_vmfault_event_reaches_health_path():
+    """End-to-end fault detection on hardware: the deliberate-OOB kernel in
+    a subprocess must surface as an amdsmi VMFAULT/page-fault event in THIS
+    process's event wait — the exact mechanism the health checker's event
+    loop runs on (the Xid-event analog)."""
+    import subprocess
+
+    import cea_amd.amdsmi as amdsmi
+    from cea_amd.amdsmi.shim import ShimAmdSmi
+
+    smi = ShimAmdSmi()
+    smi.init()
+    try:
+        smi.wait_events(10)  # arms notification on all devices
+        code = (
+            "import torch\n"
+            "from cea_amd.ops import native\n"
+            "a = torch.rand(1024, device='cuda'); b = torch.rand(1024, device='cuda')\n"
+            "c = torch.empty_like(a)\n"
+            "native.vector_add(a, b, c, inject_fault=True)\n"
+            "try: torch.cuda.synchronize()\n"
+            "except Exception: pass\n"
+        )
+        env = dict(os.environ)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        subprocess.run([sys.executable, "-c", code], env=env, timeout=240,
+                       capture_output=True)
+        deadline = time.time() + 30
+        seen = []
+        while time.time() < deadline:
+            seen += smi.wait_events(2000)
+            if any(ev.code in (1, 7) for ev in seen):  # VMFAULT / page fault
+                break
+        assert any(ev.code in (1, 7) for ev in seen), (
+            f"no VM-fault event observed; saw {[(e.code, e.message) for e in seen]}"
+        )
+    finally:
+        smi.shutdown()
+        amdsmi.ops = None
+
+
 def test_partition_status_parse_real_amdsmi():
     """amd-smi static --partition output must parse on real hardware (the
     partitioner's idempotency path)."""
