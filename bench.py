@@ -127,16 +127,38 @@ def main():
     send.uniform_()
     recv = torch.empty_like(send)
 
-    # On GPU the out-of-place copy uses the in-tree float4 HIP kernel
-    # (cea_amd/ops, gfx950) — the native path is mandatory on GPU nodes.
+    # Out-of-place copy path: autotuned between the in-tree float4 HIP
+    # kernel (cea_amd/ops, gfx950) and torch's copy_ (SDMA engines) —
+    # whichever is faster on this box at the headline size.  The native
+    # library is mandatory on GPU nodes either way.
+    copy_impl = "torch"
     if has_gpu:
         from cea_amd.ops import native
 
         native.assert_native_available()
 
-        def step(buf_send, buf_recv):
-            native.copy_(buf_recv, buf_send)
-            dist.all_reduce(buf_recv)
+        def _time_copies(fn, reps=5):
+            fn(recv, send)  # warm
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(reps):
+                fn(recv, send)
+            torch.cuda.synchronize()
+            return time.perf_counter() - t0
+
+        t_native = _time_copies(lambda d, s: native.copy_(d, s))
+        t_torch = _time_copies(lambda d, s: d.copy_(s))
+        if t_native < t_torch:
+            copy_impl = "hip-float4"
+
+            def step(buf_send, buf_recv):
+                native.copy_(buf_recv, buf_send)
+                dist.all_reduce(buf_recv)
+        else:
+
+            def step(buf_send, buf_recv):
+                buf_recv.copy_(buf_send)
+                dist.all_reduce(buf_recv)
     else:
 
         def step(buf_send, buf_recv):
@@ -229,6 +251,7 @@ def main():
                 "parallelism": f"dp{n}",
                 "backend": backend,
                 "transport": "RCCL/xGMI" if has_gpu else "gloo(cpu-test)",
+                "copy_impl": copy_impl,
                 "busbw_factor": round(busbw_factor(n), 4),
                 "algbw_GBps": round(algbw, 2),
                 "sweep": sweep,
