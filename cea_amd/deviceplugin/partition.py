@@ -10,9 +10,11 @@ devices that share a physical die are grouped by physical_index, advertised
 as `amdgpu<die>/xcd<pid>`, and a partition's DeviceSpec is just its render
 node (plus the shared /dev/kfd that the manager adds as a default device).
 
-Max-count table (parity: gpuPartitionSizeMaxCount, mig.go:36-82):
+Max-count table (parity: gpuPartitionSizeMaxCount, mig.go:36-82; the mode
+set MI355X advertises in sysfs available_compute_partition):
   SPX -> 1 partition/die (the whole GPU, 8 XCDs, 256 CUs)
   DPX -> 2 partitions/die (4 XCDs, 128 CUs each)
+  QPX -> 4 partitions/die (2 XCDs, 64 CUs each)
   CPX -> 8 partitions/die (1 XCD, 32 CUs, 36 GB HBM3E each under NPS1)
 """
 from __future__ import annotations
@@ -27,7 +29,7 @@ from ..kube import protos as api
 log = logging.getLogger(__name__)
 
 # mode -> partitions per physical die
-PARTITION_MODE_COUNT = {"SPX": 1, "DPX": 2, "CPX": 8}
+PARTITION_MODE_COUNT = {"SPX": 1, "DPX": 2, "QPX": 4, "CPX": 8}
 # accepted config spellings, e.g. "cpx", "cpx-nps1"
 def parse_partition_mode(s: str) -> str:
     mode = s.strip().upper().split("-")[0]

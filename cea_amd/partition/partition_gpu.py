@@ -27,12 +27,12 @@ log = logging.getLogger(__name__)
 
 DEFAULT_CONFIG_PATH = "/etc/amd/gpu_config.json"
 
-VALID_COMPUTE = ("SPX", "DPX", "CPX")
-VALID_MEMORY = ("NPS1", "NPS2", "NPS4")
+VALID_COMPUTE = ("SPX", "DPX", "QPX", "CPX")
+VALID_MEMORY = ("NPS1", "NPS2", "NPS4", "NPS8")
 
 # partitions per die by mode — parity with partitionSizeMaxCount
 # (partition_gpu.go:91-139)
-PARTITION_COUNT = {"SPX": 1, "DPX": 2, "CPX": 8}
+PARTITION_COUNT = {"SPX": 1, "DPX": 2, "QPX": 4, "CPX": 8}
 
 BUSY_RETRIES = 12
 BUSY_RETRY_DELAY_S = 10.0
@@ -135,8 +135,48 @@ def _set_with_busy_retry(runner: Runner, cmd: List[str]) -> None:
                          f"{BUSY_RETRIES} attempts")
 
 
+def set_compute_partition_sysfs(compute: str, sysfs_root: str = "/sys",
+                                bdfs: Optional[List[str]] = None) -> bool:
+    """Fallback setter: write the mode to the amdgpu KMD's sysfs interface
+    (/sys/class/drm/card*/device/current_compute_partition — the same knob
+    amd-smi drives).  Returns True if at least one card accepted the write.
+    Used when the amd-smi CLI refuses (e.g. library/platform mismatch) but
+    the KMD advertises the mode in available_compute_partition.
+
+    `bdfs`: restrict writes to cards whose PCI address is in the list (so a
+    tenant holding a subset of a shared host's GPUs never touches foreign
+    devices).  None = all cards (single-tenant node job)."""
+    import glob as _glob
+    import os as _os
+
+    want = {b.lower() for b in bdfs} if bdfs is not None else None
+    wrote = False
+    for card in sorted(_glob.glob(
+            _os.path.join(sysfs_root, "class", "drm", "card*", "device",
+                          "current_compute_partition"))):
+        dev_dir = _os.path.dirname(card)
+        if want is not None:
+            try:
+                pci = _os.path.basename(_os.path.realpath(dev_dir)).lower()
+            except OSError:
+                continue
+            if pci not in want:
+                continue
+        avail_path = _os.path.join(dev_dir, "available_compute_partition")
+        try:
+            with open(avail_path) as f:
+                if compute not in f.read().upper():
+                    continue
+            with open(card, "w") as f:
+                f.write(compute + "\n")
+            wrote = True
+        except OSError as e:
+            log.warning("sysfs partition write failed for %s: %s", card, e)
+    return wrote
+
+
 def run(config_path: str = DEFAULT_CONFIG_PATH,
-        runner: Runner = default_runner) -> bool:
+        runner: Runner = default_runner, sysfs_root: str = "/sys") -> bool:
     """Main flow (parity: main, partition_gpu.go:157-236).  Returns True if a
     mode change was applied, False if already in the desired state."""
     compute, memory = parse_partition_config(config_path)
@@ -159,9 +199,16 @@ def run(config_path: str = DEFAULT_CONFIG_PATH,
         _set_with_busy_retry(
             runner, ["amd-smi", "set", "--gpu", "all",
                      "--memory-partition", memory])
-    _set_with_busy_retry(
-        runner, ["amd-smi", "set", "--gpu", "all",
-                 "--compute-partition", compute])
+    try:
+        _set_with_busy_retry(
+            runner, ["amd-smi", "set", "--gpu", "all",
+                     "--compute-partition", compute])
+    except PartitionError as e:
+        # amd-smi CLI/library refusals happen on platforms whose CLI lags
+        # the KMD; fall back to the KMD sysfs knob it fronts.
+        log.warning("amd-smi set failed (%s); trying sysfs fallback", e)
+        if not set_compute_partition_sysfs(compute, sysfs_root):
+            raise
 
     states = current_partition_status(runner)
     if not check_desired(states, compute, memory):

@@ -120,13 +120,40 @@ def test_run_busy_retry(tmp_path, monkeypatch):
 def test_run_hard_failure(tmp_path):
     runner = FakeRunner([AMDSMI_SPX], set_results=[(1, "Error: not supported")])
     with pytest.raises(pg.PartitionError, match="not supported"):
-        pg.run(write_config(tmp_path, "cpx"), runner)
+        pg.run(write_config(tmp_path, "cpx"), runner,
+               sysfs_root=str(tmp_path / "sys"))
+
+
+def test_sysfs_fallback(tmp_path):
+    # amd-smi refuses, but the KMD sysfs knob accepts -> partitioning
+    # proceeds through the fallback writer
+    sysdir = tmp_path / "sys" / "class" / "drm" / "card0" / "device"
+    sysdir.mkdir(parents=True)
+    (sysdir / "available_compute_partition").write_text("SPX, DPX, QPX, CPX\n")
+    (sysdir / "current_compute_partition").write_text("SPX\n")
+    runner = FakeRunner([AMDSMI_SPX, AMDSMI_CPX_2DIES],
+                        set_results=[(1, "AMDSMI_STATUS_UNKNOWN_ERROR")])
+    assert pg.run(write_config(tmp_path, "cpx"), runner,
+                  sysfs_root=str(tmp_path / "sys")) is True
+    assert (sysdir / "current_compute_partition").read_text().strip() == "CPX"
+
+
+def test_sysfs_fallback_mode_unavailable(tmp_path):
+    sysdir = tmp_path / "sys" / "class" / "drm" / "card0" / "device"
+    sysdir.mkdir(parents=True)
+    (sysdir / "available_compute_partition").write_text("SPX\n")
+    (sysdir / "current_compute_partition").write_text("SPX\n")
+    runner = FakeRunner([AMDSMI_SPX], set_results=[(1, "unknown error")])
+    with pytest.raises(pg.PartitionError):
+        pg.run(write_config(tmp_path, "cpx"), runner,
+               sysfs_root=str(tmp_path / "sys"))
 
 
 def test_run_verification_failure(tmp_path):
     runner = FakeRunner([AMDSMI_SPX, AMDSMI_SPX])  # set "succeeds" but no change
     with pytest.raises(pg.PartitionError, match="verification failed"):
-        pg.run(write_config(tmp_path, "cpx"), runner)
+        pg.run(write_config(tmp_path, "cpx"), runner,
+               sysfs_root=str(tmp_path / "sys"))
 
 
 def test_memory_partition_change_ordered_first(tmp_path, monkeypatch):
