@@ -36,6 +36,9 @@ def parse_args(argv=None):
     p.add_argument("--gpu-fraction-divisor-file",
                    default="/etc/amd/gpu-fraction-divisor.txt")
     p.add_argument("--publish-driver-version", action="store_true")
+    p.add_argument("--mock-amdsmi", action="store_true",
+                   help="use the mock AMD-SMI backend (CPU e2e testing; the "
+                        "seam the reference swaps via nvmlutil.NvmlDeviceInfo)")
     return p.parse_args(argv)
 
 
@@ -107,7 +110,12 @@ def main(argv=None):
             log.info("waiting for amdgpu driver: %s", e)
             time.sleep(5)
 
-    smi = ShimAmdSmi()
+    if args.mock_amdsmi:
+        from cea_amd.amdsmi.mock import MockAmdSmi
+
+        smi = MockAmdSmi(args.dev_directory)
+    else:
+        smi = ShimAmdSmi()
     smi.init()   # parity: nvml.Init (nvidia_gpu.go:156)
     amdsmi.set_ops(smi)
 
