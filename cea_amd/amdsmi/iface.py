@@ -110,6 +110,12 @@ class AmdSmiOperations(ABC):
     def ecc_uncorrectable_count(self, index: int) -> int: ...
 
     @abstractmethod
+    def xgmi_error_status(self, index: int) -> int:
+        """xGMI link error state: 0 = none, 1 = errors, 2 = multiple
+        (amdsmi_xgmi_status_t).  Polled by the health watchdog for the
+        synthetic code-63 event."""
+
+    @abstractmethod
     def driver_version(self) -> str:
         """amdgpu KMD version string, e.g. '6.10.5'."""
 

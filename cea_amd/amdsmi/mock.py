@@ -43,6 +43,7 @@ class MockAmdSmi(AmdSmiOperations):
         self.inited = False
         self._events: "queue.Queue[Event]" = queue.Queue()
         self._ecc_counts = {}
+        self._xgmi_status = {}
         self._lock = threading.Lock()
 
     # -- discovery ----------------------------------------------------------
@@ -102,6 +103,10 @@ class MockAmdSmi(AmdSmiOperations):
         with self._lock:
             return self._ecc_counts.get(index, 0)
 
+    def xgmi_error_status(self, index: int) -> int:
+        with self._lock:
+            return self._xgmi_status.get(index, 0)
+
     def driver_version(self) -> str:
         return "6.10.5"
 
@@ -127,6 +132,10 @@ class MockAmdSmi(AmdSmiOperations):
     def set_ecc_count(self, index: int, count: int) -> None:
         with self._lock:
             self._ecc_counts[index] = count
+
+    def set_xgmi_status(self, index: int, status: int) -> None:
+        with self._lock:
+            self._xgmi_status[index] = status
 
 
 def make_fake_dev(dev_dir: str, num_gpus: int, first_minor: int = 128) -> None:

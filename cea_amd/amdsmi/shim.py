@@ -163,6 +163,14 @@ class ShimAmdSmi(AmdSmiOperations):
         )
         return uncorr.value
 
+    def xgmi_error_status(self, index: int) -> int:
+        status = ctypes.c_int()
+        self._check(
+            self.lib.cea_smi_xgmi_error_status(index, ctypes.byref(status)),
+            f"xgmi_error_status({index})",
+        )
+        return status.value
+
     def driver_version(self) -> str:
         buf = ctypes.create_string_buffer(256)
         self._check(self.lib.cea_smi_driver_version(buf, 256), "driver_version")

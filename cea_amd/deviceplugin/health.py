@@ -34,6 +34,7 @@ from .. import amdsmi
 from ..amdsmi.iface import (
     EVT_ECC_UNCORRECTABLE,
     EVT_LOST,
+    EVT_XGMI_ERROR,
     MONITOR_CRITICAL_EVENTS,
     Event,
 )
@@ -85,6 +86,7 @@ class GPUHealthChecker:
         self._stop = threading.Event()
         self._threads = []
         self._ecc_baseline: Dict[int, int] = {}
+        self._xgmi_state: Dict[int, int] = {}
         self._condition_lock = threading.Lock()
 
     # -- lifecycle -----------------------------------------------------------
@@ -158,6 +160,21 @@ class GPUHealthChecker:
                         device_uuid=uuid, code=EVT_ECC_UNCORRECTABLE,
                         message=f"uncorrectable ECC count {count}",
                     ))
+                # xGMI link errors: synthetic code 63 on 0->error transition
+                try:
+                    xgmi = ops.xgmi_error_status(i)
+                except Exception:  # noqa: BLE001 - not supported on 1-GPU
+                    xgmi = 0
+                if xgmi != 0 and self._xgmi_state.get(i, 0) == 0:
+                    try:
+                        uuid = ops.device_info(i).uuid
+                    except Exception:  # noqa: BLE001
+                        uuid = ""
+                    self.catch_error(Event(
+                        device_uuid=uuid, code=EVT_XGMI_ERROR,
+                        message=f"xGMI link error status {xgmi}",
+                    ))
+                self._xgmi_state[i] = xgmi
 
     # -- the policy core ------------------------------------------------------
     def catch_error(self, ev: Event) -> None:

@@ -356,6 +356,23 @@ int cea_smi_ecc_count(int idx, unsigned long long* correctable,
   return 0;
 }
 
+// xGMI link error state: 0 = no errors, 1 = errors, 2 = multiple errors
+// (amdsmi_xgmi_status_t).  The health checker's polling watchdog raises
+// synthetic event 63 on nonzero (parity: Xid 63 class link errors).
+int cea_smi_xgmi_error_status(int idx, int* status) {
+  amdsmi_processor_handle h;
+  {
+    std::lock_guard<std::mutex> lk(g_mutex);
+    if (idx < 0 || idx >= (int)g_devices.size()) return -2;
+    h = g_devices[idx].handle;
+  }
+  amdsmi_xgmi_status_t st{};
+  amdsmi_status_t rc = amdsmi_gpu_xgmi_error_status(h, &st);
+  if (rc != AMDSMI_STATUS_SUCCESS) return fail("xgmi_error_status", rc);
+  *status = (int)st;
+  return 0;
+}
+
 int cea_smi_driver_version(char* buf, int len) {
   amdsmi_processor_handle h;
   {

@@ -184,3 +184,31 @@ def test_event_loop_consumes_injected_events(tmp_path):
         assert [d.ID for d in found] == ["amdgpu0"]
     finally:
         hc.stop()
+
+
+def test_xgmi_error_poll_raises_code_63(tmp_path):
+    from cea_amd.amdsmi.iface import EVT_XGMI_ERROR
+
+    mgr, mock, kube, hc = make_env(tmp_path)
+    hc.ecc_poll_interval_s = 0.05
+    hc.start()
+    try:
+        import time
+        mock.set_xgmi_status(0, 1)
+        deadline = time.time() + 3
+        cond = None
+        while time.time() < deadline and cond is None:
+            conds = kube.nodes["node1"]["status"]["conditions"]
+            cond = next((c for c in conds if c["type"] == CONDITION_TYPE), None)
+            time.sleep(0.05)
+        assert cond is not None, "xGMI error never reached the node condition"
+        assert EVT_XGMI_ERROR in json.loads(cond["reason"])
+        # monitor-only by default: no device goes Unhealthy
+        assert drain(mgr.health) == []
+        # only raised once per 0->error transition
+        time.sleep(0.3)
+        n_events = len(kube.events)
+        time.sleep(0.3)
+        assert len(kube.events) == n_events
+    finally:
+        hc.stop()
