@@ -48,3 +48,13 @@ presubmit: lint all test
 
 clean:
 	rm -f $(SMI_SO) $(GPU_SO) $(RCCL_BENCH)
+
+# container image targets — parity /root/reference/Makefile:46-93
+IMAGE ?= cea-amd/gpu-device-plugin
+TAG ?= $(shell cat VERSION 2>/dev/null || echo dev)
+
+image:
+	docker build -t $(IMAGE):$(TAG) .
+
+push: image
+	docker push $(IMAGE):$(TAG)
