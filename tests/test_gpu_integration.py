@@ -83,6 +83,39 @@ def test_metrics_collector_on_hardware():
         amdsmi.ops = None
 
 
+def test_windowed_utilization_rises_under_load():
+    """The shim's C++ sampler must report elevated GFX busy% while a
+    compute loop runs — validates the windowed-average path the metrics
+    server publishes (the nvmlDeviceGetAverageUsage analog)."""
+    import threading
+
+    import cea_amd.amdsmi as amdsmi
+    from cea_amd.amdsmi.shim import ShimAmdSmi
+
+    smi = ShimAmdSmi(sampler_interval_ms=100)
+    smi.init()
+    try:
+        stop = threading.Event()
+
+        def burn():
+            a = torch.randn(4096, 4096, device="cuda")
+            while not stop.is_set():
+                a = a @ a
+                a = a / a.norm()
+            torch.cuda.synchronize()
+
+        t = threading.Thread(target=burn, daemon=True)
+        t.start()
+        time.sleep(4)  # accumulate ~40 samples under load
+        busy = smi.average_gfx_utilization(0, 3.0)
+        stop.set()
+        t.join(timeout=30)
+        assert busy > 30, f"GFX busy {busy}% under a matmul loop"
+    finally:
+        smi.shutdown()
+        amdsmi.ops = None
+
+
 def test_vmfault_event_reaches_health_path():
     """End-to-end fault detection on hardware: the deliberate-OOB kernel in
     a subprocess must surface as an amdsmi VMFAULT/page-fault event in THIS
