@@ -61,6 +61,77 @@ def test_plugin_end_to_end_on_hardware(tmp_path):
         stub.stop()
 
 
+def test_fault_marks_device_unhealthy_end_to_end(tmp_path):
+    """The full production loop on hardware: plugin with health monitoring
+    (EVENT_CONFIG=1 makes VM faults health-critical), a pod-like subprocess
+    triggers the OOB kernel, and the kubelet-side ListAndWatch stream must
+    receive a resend with the device Unhealthy."""
+    import subprocess
+    import threading
+
+    plugin_dir = str(tmp_path / "plugin")
+    os.makedirs(plugin_dir)
+    stub = KubeletStub(plugin_dir)
+    stub.start()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["EVENT_CONFIG"] = "1,7"   # VMFAULT / page-fault => health-critical
+    env.pop("NODE_NAME", None)    # no kube client in this tier
+    proc = subprocess.Popen(
+        [sys.executable, os.path.join(REPO, "cmd", "amd_gpu.py"),
+         "--plugin-directory", plugin_dir,
+         "--enable-health-monitoring",
+         "--gpu-config", str(tmp_path / "missing.json")],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        assert stub.registered.wait(60)
+        client = PluginClient(os.path.join(plugin_dir, "amdgpu.sock"))
+        stream = client.list_and_watch_once(timeout=240)
+        it = iter(stream)
+        first = next(it)
+        assert all(d.health == "Healthy" for d in first.devices)
+
+        # fire the fault from a separate process (a "pod")
+        code = (
+            "import torch\n"
+            "from cea_amd.ops import native\n"
+            "a = torch.rand(1024, device='cuda'); b = torch.rand(1024, device='cuda')\n"
+            "c = torch.empty_like(a)\n"
+            "native.vector_add(a, b, c, inject_fault=True)\n"
+            "try: torch.cuda.synchronize()\n"
+            "except Exception: pass\n"
+        )
+        subprocess.run([sys.executable, "-c", code], env=env, timeout=240,
+                       capture_output=True)
+
+        # the health checker's 5s event wait must push an Unhealthy resend
+        got_unhealthy = threading.Event()
+
+        def reader():
+            try:
+                for resp in it:
+                    if any(d.health == "Unhealthy" for d in resp.devices):
+                        got_unhealthy.set()
+                        return
+            except Exception:  # noqa: BLE001 - stream cancel
+                pass
+
+        t = threading.Thread(target=reader, daemon=True)
+        t.start()
+        assert got_unhealthy.wait(60), \
+            "fault did not propagate to ListAndWatch as Unhealthy"
+        stream.cancel()
+        client.close()
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        stub.stop()
+
+
 def test_metrics_collector_on_hardware():
     import cea_amd.amdsmi as amdsmi
     from cea_amd.amdsmi.shim import ShimAmdSmi
