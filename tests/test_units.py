@@ -219,3 +219,16 @@ def test_parse_gpu_config_and_divisor(tmp_path):
     div.write_text("0")
     assert amd_gpu.parse_gpu_fraction_divisor(str(div)) == 1
     assert amd_gpu.parse_gpu_fraction_divisor(str(tmp_path / "nope")) == 1
+
+
+def test_shell_scripts_parse():
+    """Every installer/demo shell script must pass `bash -n`."""
+    import glob
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    scripts = glob.glob(os.path.join(repo, "amd-driver-installer", "**", "*.sh"),
+                        recursive=True) + \
+        glob.glob(os.path.join(repo, "demo", "**", "*.sh"), recursive=True)
+    assert len(scripts) >= 3, scripts
+    for s in scripts:
+        subprocess.run(["bash", "-n", s], check=True)
