@@ -149,7 +149,7 @@ def main():
         t_native = _time_copies(lambda d, s: native.copy_(d, s))
         t_torch = _time_copies(lambda d, s: d.copy_(s))
         if t_native < t_torch:
-            copy_impl = "hip-float4"
+            copy_impl = "hip-exact-nt"
 
             def step(buf_send, buf_recv):
                 native.copy_(buf_recv, buf_send)
@@ -169,6 +169,33 @@ def main():
         if has_gpu:
             torch.cuda.synchronize()
 
+    # hipGraph capture of the step: one replay per step instead of two eager
+    # launches — removes per-launch CPU overhead, which dominates the small
+    # end of the sweep (a 1 MiB step is ~13 us eager, most of it launch).
+    # RCCL collectives are capturable (comm bound via device_id above);
+    # falls back to eager if capture fails.
+    use_graphs = has_gpu and os.environ.get("CEA_BENCH_GRAPH", "1") == "1"
+
+    def make_step(buf_send, buf_recv):
+        """Returns (callable, is_graph) executing one out-of-place step."""
+        if not use_graphs:
+            return (lambda: step(buf_send, buf_recv)), False
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    step(buf_send, buf_recv)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                step(buf_send, buf_recv)
+            return g.replay, True
+        except Exception:  # noqa: BLE001 - capture support is optional
+            torch.cuda.synchronize()
+            return (lambda: step(buf_send, buf_recv)), False
+
     def busbw_factor(nranks: int) -> float:
         return 2.0 * (nranks - 1) / nranks if nranks > 1 else 1.0
 
@@ -179,13 +206,14 @@ def main():
         while size <= args.max_bytes:
             e = size // 4
             s_send, s_recv = send[:e], recv[:e]
+            run_step, _ = make_step(s_send, s_recv)
             for _ in range(args.sweep_warmup):
-                step(s_send, s_recv)
+                run_step()
             sync()
             dist.barrier()
             t0 = time.perf_counter()
             for _ in range(args.sweep_iters):
-                step(s_send, s_recv)
+                run_step()
             sync()
             el = time.perf_counter() - t0
             dist.barrier()
@@ -204,14 +232,15 @@ def main():
             size *= 2
 
     # ---- timed region: K steps at the headline size ----------------------
+    run_step, is_graph = make_step(send, recv)
     for _ in range(args.warmup):
-        step(send, recv)
+        run_step()
     sync()
     dist.barrier()
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step(send, recv)
+        run_step()
     sync()
     elapsed = time.perf_counter() - t0
     dist.barrier()
@@ -252,6 +281,7 @@ def main():
                 "backend": backend,
                 "transport": "RCCL/xGMI" if has_gpu else "gloo(cpu-test)",
                 "copy_impl": copy_impl,
+                "hip_graph": bool(is_graph),
                 "busbw_factor": round(busbw_factor(n), 4),
                 "algbw_GBps": round(algbw, 2),
                 "sweep": sweep,

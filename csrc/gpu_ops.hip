@@ -66,11 +66,27 @@ __global__ void vector_add_tail(const float* a, const float* b, float* c,
   if (i < n) c[i] = a[i] + b[i];
 }
 
-__global__ void copy_vec4(const float4* __restrict__ src,
-                          float4* __restrict__ dst, long n4) {
+// Native 16-B vector type: HIP's float4 is a class, and the nontemporal
+// builtins only take real vector types.
+typedef float vf4 __attribute__((ext_vector_type(4)));
+
+// Exact-cover copy: one 16-B vector per lane, no loop.  Tuned on MI355X
+// (profiles/copy_tune_r01.csv): beats every grid-stride geometry.
+__global__ void copy_exact(const vf4* __restrict__ src,
+                           vf4* __restrict__ dst, long n4) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (; i < n4; i += stride) dst[i] = src[i];
+  if (i < n4) dst[i] = src[i];
+}
+
+// Nontemporal variant for streaming sizes (working set past the 256 MiB
+// Infinity Cache): sc-bit loads/stores keep the one-shot traffic from
+// displacing L2/L3 lines — 6.56 TB/s r+w at 512 MiB vs 4.9-5.5 for the
+// grid-stride cached copy.
+__global__ void copy_exact_nt(const vf4* __restrict__ src,
+                              vf4* __restrict__ dst, long n4) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n4)
+    __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
 }
 
 // Wave64 shuffle reduction -> per-block LDS reduction -> one atomic/block.
@@ -166,19 +182,38 @@ int cea_vector_add(const float* a, const float* b, float* c, long n,
   return 0;
 }
 
+namespace {
+
+// Tuned dispatch (profiles/copy_tune_r01.csv, MI355X): while src+dst still
+// fit the 256 MiB Infinity Cache a plain cached copy wins (~6.9 TB/s r+w at
+// 64 MiB); past it the nontemporal exact-cover kernel wins (~6.56 TB/s at
+// 512 MiB vs ~4.9 for grid-stride cached).
+void launch_copy(const void* src, void* dst, long n4, hipStream_t s) {
+  long bytes = n4 * 16;
+  if (2 * bytes <= (256L << 20)) {
+    const int block = 512;
+    long g = (n4 + block - 1) / block;
+    hipLaunchKernelGGL(copy_exact, dim3((unsigned)(g ? g : 1)), dim3(block), 0,
+                       s, (const vf4*)src, (vf4*)dst, n4);
+  } else {
+    const int block = 256;
+    long g = (n4 + block - 1) / block;
+    hipLaunchKernelGGL(copy_exact_nt, dim3((unsigned)g), dim3(block), 0, s,
+                       (const vf4*)src, (vf4*)dst, n4);
+  }
+}
+
+}  // namespace
+
 // Plain async d2d copy of `bytes` bytes (multiple of 16) on `stream` —
-// used by bench.py's out-of-place all_reduce step (float4-vectorized, fills
-// the 256-CU chip; measured ~5.5 TB/s read+write on MI355X vs ~5.1 for
-// torch copy_).
+// used by bench.py's out-of-place all_reduce step.  Measured 6.9 TB/s r+w
+// L3-resident / 6.56 TB/s streaming on MI355X vs ~5.1 for torch copy_.
 int cea_copy(void* dst, const void* src, long bytes, void* stream) {
   if (bytes % 16) {
     snprintf(g_err, sizeof(g_err), "bytes must be a multiple of 16");
     return -2;
   }
-  hipStream_t s = (hipStream_t)stream;
-  long n4 = bytes / 16;
-  hipLaunchKernelGGL(copy_vec4, grid_for(n4, 256), dim3(256), 0, s,
-                     (const float4*)src, (float4*)dst, n4);
+  launch_copy(src, dst, bytes / 16, (hipStream_t)stream);
   CHECK(hipGetLastError());
   return 0;
 }
@@ -194,18 +229,14 @@ int cea_copy_bw(void* dst, const void* src, long bytes, int iters,
   }
   hipStream_t s = (hipStream_t)stream;
   long n4 = bytes / 16;
-  const int block = 256;
   hipEvent_t t0, t1;
   CHECK(hipEventCreate(&t0));
   CHECK(hipEventCreate(&t1));
   // warmup
-  hipLaunchKernelGGL(copy_vec4, grid_for(n4, block), dim3(block), 0, s,
-                     (const float4*)src, (float4*)dst, n4);
+  launch_copy(src, dst, n4, s);
   CHECK(hipGetLastError());
   CHECK(hipEventRecord(t0, s));
-  for (int i = 0; i < iters; i++)
-    hipLaunchKernelGGL(copy_vec4, grid_for(n4, block), dim3(block), 0, s,
-                       (const float4*)src, (float4*)dst, n4);
+  for (int i = 0; i < iters; i++) launch_copy(src, dst, n4, s);
   CHECK(hipEventRecord(t1, s));
   CHECK(hipEventSynchronize(t1));
   float ms = 0.f;
