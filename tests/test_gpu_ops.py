@@ -106,3 +106,18 @@ def test_amdsmi_shim_on_hardware():
         assert ver
     finally:
         smi.shutdown()
+
+
+def test_copy_numerics_both_dispatch_branches():
+    """cea_copy dispatches cached exact-cover (<=128 MiB) vs nontemporal
+    (>128 MiB); both must be byte-exact, including ragged tails."""
+    for elems in (
+        (64 << 20) // 4,        # 64 MiB: cached branch
+        (160 << 20) // 4,       # 160 MiB: nontemporal branch
+        (64 << 20) // 4 + 4,    # ragged: not a multiple of block*16B
+    ):
+        src = torch.rand(elems, device="cuda")
+        dst = torch.empty_like(src)
+        native.copy_(dst, src)
+        torch.cuda.synchronize()
+        assert torch.equal(dst, src), elems

@@ -232,3 +232,30 @@ def test_shell_scripts_parse():
     assert len(scripts) >= 3, scripts
     for s in scripts:
         subprocess.run(["bash", "-n", s], check=True)
+
+
+def test_gpu_readiness_helpers(tmp_path):
+    """Readiness gate: devices_ready needs kfd + renderD* + amdgpu module;
+    confidential type file is optional (persistenced-installer parity)."""
+    import importlib.util
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        "gpu_readiness", os.path.join(repo, "cmd", "gpu_readiness.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+
+    dev = tmp_path / "dev"
+    sysfs = tmp_path / "sys"
+    (dev / "dri").mkdir(parents=True)
+    (sysfs / "module").mkdir(parents=True)
+    assert not mod.devices_ready(str(dev), str(sysfs))
+    (dev / "kfd").touch()
+    (dev / "dri" / "renderD128").touch()
+    assert not mod.devices_ready(str(dev), str(sysfs))  # module missing
+    (sysfs / "module" / "amdgpu").mkdir()
+    assert mod.devices_ready(str(dev), str(sysfs))
+
+    assert mod.confidential_node_type(str(tmp_path / "absent")) == ""
+    f = tmp_path / "ctype.txt"
+    f.write_text("SEV-SNP\n")
+    assert mod.confidential_node_type(str(f)) == "sev-snp"
