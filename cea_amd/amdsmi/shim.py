@@ -62,9 +62,13 @@ def load_library() -> ctypes.CDLL:
 
 
 class ShimAmdSmi(AmdSmiOperations):
-    # amdsmi event-notification mask: VMFAULT|THERMAL|PRE_RESET|POST_RESET|
-    # RING_HANG (AMDSMI_EVENT_MASK_FROM_INDEX(i) = 1<<(i-1))
-    DEFAULT_EVENT_MASK = (1 << 0) | (1 << 1) | (1 << 2) | (1 << 3) | (1 << 4)
+    # amdsmi event-notification mask (AMDSMI_EVENT_MASK_FROM_INDEX(i) =
+    # 1<<(i-1)): VMFAULT(1) | THERMAL_THROTTLE(2) | GPU_PRE_RESET(3) |
+    # GPU_POST_RESET(4) | PAGE_FAULT_START(7) | PAGE_FAULT_END(8) — the
+    # health-relevant subset of amdsmi_evt_notification_type_t.
+    DEFAULT_EVENT_MASK = (
+        (1 << 0) | (1 << 1) | (1 << 2) | (1 << 3) | (1 << 6) | (1 << 7)
+    )
 
     def __init__(self, sampler_interval_ms: int = 160):
         self.lib = load_library()
