@@ -345,11 +345,17 @@ class AmdGPUManager:
             self.grpc_server = server
             log.info("device plugin listening on %s", socket_path)
             try:
+                # Snapshot the kubelet-socket identity BEFORE the
+                # registration attempt: a kubelet that comes up between a
+                # failed registration and the watchdog's first stat would
+                # otherwise be baselined as "already there" and never
+                # trigger the re-register restart.
+                kubelet_watch = FileWatcher(kubelet_socket)
                 if os.path.exists(kubelet_socket):
                     self.register_with_kubelet(kubelet_socket)
                 else:
                     log.warning("kubelet socket %s absent; serving unregistered", kubelet_socket)
-                self._status_check_loop(socket_path, kubelet_socket)
+                self._status_check_loop(socket_path, kubelet_watch)
             finally:
                 server.stop(grace=0.5)
                 self.grpc_server = None
@@ -375,8 +381,8 @@ class AmdGPUManager:
             )
         log.info("registered %s with kubelet", RESOURCE_NAME)
 
-    def _status_check_loop(self, socket_path: str, kubelet_socket: str) -> None:
-        kubelet_watch = FileWatcher(kubelet_socket)
+    def _status_check_loop(self, socket_path: str,
+                           kubelet_watch: FileWatcher) -> None:
         last_gpu_check = time.monotonic()
         while not self._stop.is_set():
             time.sleep(SOCKET_CHECK_INTERVAL_S)
