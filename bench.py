@@ -8,11 +8,12 @@ sweep 1 MiB -> 512 MiB, factor 2, 5 warmup, 100 iters, no data check
 the same protocol through torch.distributed (backend "nccl" IS RCCL on
 ROCm), one process per GPU over xGMI.
 
-Step semantics (the driver's timed contract): one step = one out-of-place
-all_reduce of the headline 512 MiB fp32 buffer (copy into the recv buffer +
-in-place collective — the same work rccl-tests' out-of-place mode times).
-The full message sweep runs outside the timed region and is reported in
-config.sweep for the scaling curve.
+Step semantics (the driver's timed contract): one step = one all_reduce of
+the headline 512 MiB fp32 buffer.  For world>=2 it is the in-place
+collective — exactly what rccl-tests' in-place row times; for world==1
+(where all_reduce is a no-op) it is the out-of-place d2d copy a world-1
+out-of-place all_reduce performs.  The full message sweep runs outside the
+timed region and is reported in config.sweep for the scaling curve.
 
 bus-bw convention (nccl-tests): busbw = algbw * 2*(n-1)/n for all_reduce.
 For n==1 that factor degenerates to 0, so the single-GPU value reported is
@@ -127,42 +128,58 @@ def main():
     send.uniform_()
     recv = torch.empty_like(send)
 
-    # Out-of-place copy path: autotuned between the in-tree float4 HIP
-    # kernel (cea_amd/ops, gfx950) and torch's copy_ (SDMA engines) —
-    # whichever is faster on this box at the headline size.  The native
-    # library is mandatory on GPU nodes either way.
-    copy_impl = "torch"
-    if has_gpu:
-        from cea_amd.ops import native
+    # Step protocol (rccl-tests parity):
+    #   n >= 2 — in-place all_reduce, exactly what rccl-tests' in-place row
+    #     times (one ncclAllReduce, no extra traffic).  torch's all_reduce
+    #     is in-place-only, so the out-of-place emulation (copy + reduce)
+    #     would charge a send->recv copy rccl-tests does not time.
+    #   n == 1 — all_reduce degenerates to a no-op, so time the d2d copy a
+    #     world-1 out-of-place all_reduce performs, autotuned between the
+    #     in-tree gfx950 kernel (cea_amd/ops) and torch copy_ (SDMA).
+    copy_impl = None
+    mode = "in-place"
+    if n == 1:
+        mode = "out-of-place"
+        copy_impl = "torch"
+        if has_gpu:
+            from cea_amd.ops import native
 
-        native.assert_native_available()
+            native.assert_native_available()
 
-        def _time_copies(fn, reps=5):
-            fn(recv, send)  # warm
-            torch.cuda.synchronize()
-            t0 = time.perf_counter()
-            for _ in range(reps):
-                fn(recv, send)
-            torch.cuda.synchronize()
-            return time.perf_counter() - t0
+            def _time_copies(fn, reps=5):
+                fn(recv, send)  # warm
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(reps):
+                    fn(recv, send)
+                torch.cuda.synchronize()
+                return time.perf_counter() - t0
 
-        t_native = _time_copies(lambda d, s: native.copy_(d, s))
-        t_torch = _time_copies(lambda d, s: d.copy_(s))
-        if t_native < t_torch:
-            copy_impl = "hip-exact-nt"
+            t_native = _time_copies(lambda d, s: native.copy_(d, s))
+            t_torch = _time_copies(lambda d, s: d.copy_(s))
+            if t_native < t_torch:
+                copy_impl = "hip-exact-nt"
 
-            def step(buf_send, buf_recv):
-                native.copy_(buf_recv, buf_send)
-                dist.all_reduce(buf_recv)
+                def step(buf_send, buf_recv):
+                    native.copy_(buf_recv, buf_send)
+                    dist.all_reduce(buf_recv)
+            else:
+
+                def step(buf_send, buf_recv):
+                    buf_recv.copy_(buf_send)
+                    dist.all_reduce(buf_recv)
         else:
 
             def step(buf_send, buf_recv):
                 buf_recv.copy_(buf_send)
                 dist.all_reduce(buf_recv)
     else:
+        if has_gpu:
+            from cea_amd.ops import native
+
+            native.assert_native_available()
 
         def step(buf_send, buf_recv):
-            buf_recv.copy_(buf_send)
             dist.all_reduce(buf_recv)
 
     def sync():
@@ -177,7 +194,7 @@ def main():
     use_graphs = has_gpu and os.environ.get("CEA_BENCH_GRAPH", "1") == "1"
 
     def make_step(buf_send, buf_recv):
-        """Returns (callable, is_graph) executing one out-of-place step."""
+        """Returns (callable, is_graph) executing one timed step."""
         if not use_graphs:
             return (lambda: step(buf_send, buf_recv)), False
         try:
@@ -274,7 +291,7 @@ def main():
                 "model": "all_reduce_perf",
                 "collective": "all_reduce",
                 "message_bytes": args.max_bytes,
-                "mode": "out-of-place",
+                "mode": mode,
                 "global_batch": None,
                 "seq_len": None,
                 "parallelism": f"dp{n}",
