@@ -15,7 +15,8 @@ def load_all():
             glob.glob(os.path.join(REPO, "amd-driver-installer", "**", "*.yaml"),
                       recursive=True) + \
             glob.glob(os.path.join(REPO, "demo", "**", "*.yaml"), recursive=True) + \
-            glob.glob(os.path.join(REPO, "example", "**", "*.yaml"), recursive=True):
+            glob.glob(os.path.join(REPO, "example", "**", "*.yaml"), recursive=True) + \
+            [os.path.join(REPO, "daemonset.yaml")]:
         with open(path) as f:
             docs[path] = [d for d in yaml.safe_load_all(f) if d]
     return docs
