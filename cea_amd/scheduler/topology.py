@@ -113,8 +113,21 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
 
 
 def _pairwise_score(topos: List[TopoKey]) -> int:
-    score = 0
-    for i in range(len(topos)):
-        for j in range(i + 1, len(topos)):
-            score += distance(topos[i], topos[j])
+    """Summed pairwise distance in O(G) instead of O(G^2): distance terms
+    are label-equality indicators, so sum over pairs = weight x (total
+    pairs - same-label pairs), with same-label pairs counted per group."""
+    g = len(topos)
+    total_pairs = g * (g - 1) // 2
+
+    def same_pairs(keyfn) -> int:
+        counts: dict = {}
+        for t in topos:
+            k = keyfn(t)
+            counts[k] = counts.get(k, 0) + 1
+        return sum(c * (c - 1) // 2 for c in counts.values())
+
+    # Labels compared exactly as distance() does (bare label equality).
+    score = W_BLOCK * (total_pairs - same_pairs(lambda t: t.block))
+    score += W_SUBBLOCK * (total_pairs - same_pairs(lambda t: t.subblock))
+    score += W_HOST * (total_pairs - same_pairs(lambda t: t.host))
     return score

@@ -259,3 +259,42 @@ def test_partition_status_parse_real_amdsmi():
     if states[0]["accelerator_partition"] == "SPX":
         assert check_desired(states, "SPX",
                              states[0].get("memory_partition", "NPS1"))
+
+
+def test_hsa_cu_mask_fencing_throughput():
+    """cu-fencing (the MPS-analog sharing strategy) must actually fence
+    compute: a compute-bound GEMM under the manager's HSA_CU_MASK env for a
+    1-of-8 client (32 of 256 CUs) must run much slower than unmasked.
+    Parity role: example/cuda-mps/cuda_mem_and_sm_count.c verifying MPS
+    thread fencing."""
+    import subprocess
+
+    code = (
+        "import torch, time\n"
+        "a = torch.randn(4096, 4096, device='cuda')\n"
+        "b = torch.randn(4096, 4096, device='cuda')\n"
+        "for _ in range(3): (a @ b)\n"
+        "torch.cuda.synchronize()\n"
+        "t0 = time.perf_counter()\n"
+        "for _ in range(20): (a @ b)\n"
+        "torch.cuda.synchronize()\n"
+        "print(time.perf_counter() - t0)\n"
+    )
+
+    def run(mask_env):
+        env = dict(os.environ)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        env.update(mask_env)
+        r = subprocess.run([sys.executable, "-c", code], env=env,
+                           timeout=240, capture_output=True, text=True)
+        assert r.returncode == 0, r.stderr[-800:]
+        return float(r.stdout.strip().splitlines()[-1])
+
+    t_full = run({})
+    # the manager's envs() emits HSA_CU_MASK=0:0-<n> for cu-fencing clients;
+    # 32 CUs = one 1-of-8 share of the 256-CU die
+    t_fenced = run({"HSA_CU_MASK": "0:0-31"})
+    assert t_fenced > 2.0 * t_full, (
+        f"CU mask did not fence compute: full={t_full:.3f}s fenced={t_fenced:.3f}s"
+    )

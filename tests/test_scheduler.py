@@ -201,3 +201,41 @@ def test_labeler():
     label_node(kube, "n1", {"block": "b", "subblock": "s", "host": "h"})
     assert kube.nodes["n1"]["metadata"]["labels"][
         "topology.cea-amd.io/block"] == "b"
+
+
+def test_assign_scales_to_large_clusters():
+    """Control-plane perf: a 512-pod gang over 2048 nodes must place in
+    seconds (the reference's exhaustive combinatorial search,
+    schedule-daemon.py:500-544, cannot) and must still pick one packed
+    block when one exists."""
+    import time as _time
+
+    nodes = []
+    # 256 blocks x 8 hosts, 1 pod capacity each; block 77 has 64 hosts
+    # with capacity 8 = the only single-block fit for 512 pods.
+    for b in range(256):
+        for h in range(8):
+            nodes.append(CandidateNode(
+                f"n-{b}-{h}", K(f"b{b:03d}", "s0", f"h{h}"), 1))
+    for h in range(64):
+        nodes.append(CandidateNode(
+            f"big-{h}", K("b077x", "s0", f"H{h}"), 8))
+    t0 = _time.perf_counter()
+    got = assign_pods(512, nodes)
+    elapsed = _time.perf_counter() - t0
+    assert got is not None and len(got) == 512
+    assert elapsed < 10.0, f"placement took {elapsed:.1f}s"
+    assert all(n.startswith("big-") for n in got), \
+        "did not pick the packed block"
+
+
+def test_pairwise_score_matches_bruteforce():
+    import random
+    rng = random.Random(7)
+    for _ in range(20):
+        topos = [K(f"b{rng.randint(0, 3)}", f"s{rng.randint(0, 2)}",
+                   f"h{rng.randint(0, 4)}") for _ in range(rng.randint(1, 12))]
+        brute = sum(topo.distance(topos[i], topos[j])
+                    for i in range(len(topos))
+                    for j in range(i + 1, len(topos)))
+        assert topo._pairwise_score(topos) == brute
