@@ -54,6 +54,8 @@ struct Args {
   int warmup = 5;                  // -w
   int iters = 100;                 // -n / --iters
   int check = 0;                   // -c
+  int graph = 0;                   // -G: capture the op in a hipGraph and
+                                   //     time graph replays (nccl-tests -G)
   std::string op = "all_reduce";   // -o
 };
 
@@ -135,10 +137,11 @@ int main(int argc, char** argv) {
     else if (f == "-w") a.warmup = atoi(next());
     else if (f == "-n" || f == "--iters") a.iters = atoi(next());
     else if (f == "-c") a.check = atoi(next());
+    else if (f == "-G") a.graph = atoi(next());
     else if (f == "-o") a.op = next();
     else if (f == "-h" || f == "--help") {
       printf("usage: %s [-b min] [-e max] [-f factor] [-g ngpus] [-w warmup]"
-             " [-n iters] [-c check] [-o all_reduce|all_gather|"
+             " [-n iters] [-c check] [-G graph] [-o all_reduce|all_gather|"
              "reduce_scatter|broadcast]\n", argv[0]);
       return 0;
     }
@@ -181,12 +184,45 @@ int main(int argc, char** argv) {
 
   for (long bytes = a.min_bytes; bytes <= a.max_bytes; bytes *= a.factor) {
     long count = bytes / 4;
-    for (int w = 0; w < a.warmup; w++) run_op(c, a.op, count);
+
+    // -G: capture one grouped op into a hipGraph per stream, then time
+    // whole-graph replays (one replay = one collective) — removes the
+    // per-iteration launch overhead that dominates small messages.
+    std::vector<hipGraphExec_t> execs;
+    if (a.graph > 0) {
+      for (int w = 0; w < a.warmup; w++) run_op(c, a.op, count);  // eager warm
+      sync_all(c);
+      for (int i = 0; i < c.n; i++) {
+        HIPCHECK(hipSetDevice(i));
+        HIPCHECK(hipStreamBeginCapture(c.streams[i],
+                                       hipStreamCaptureModeRelaxed));
+      }
+      run_op(c, a.op, count);
+      execs.resize(c.n);
+      for (int i = 0; i < c.n; i++) {
+        hipGraph_t g;
+        HIPCHECK(hipSetDevice(i));
+        HIPCHECK(hipStreamEndCapture(c.streams[i], &g));
+        HIPCHECK(hipGraphInstantiate(&execs[i], g, nullptr, nullptr, 0));
+        HIPCHECK(hipGraphDestroy(g));
+      }
+    }
+    auto launch = [&]() {
+      if (a.graph > 0) {
+        for (int i = 0; i < c.n; i++)
+          HIPCHECK(hipGraphLaunch(execs[i], c.streams[i]));
+      } else {
+        run_op(c, a.op, count);
+      }
+    };
+
+    for (int w = 0; w < a.warmup; w++) launch();
     sync_all(c);
     auto t0 = std::chrono::steady_clock::now();
-    for (int it = 0; it < a.iters; it++) run_op(c, a.op, count);
+    for (int it = 0; it < a.iters; it++) launch();
     sync_all(c);
     auto t1 = std::chrono::steady_clock::now();
+    for (auto& ge : execs) HIPCHECK(hipGraphExecDestroy(ge));
     double us =
         std::chrono::duration<double, std::micro>(t1 - t0).count() / a.iters;
     double algbw = bytes / us / 1e3;  // bytes/us -> GB/s
