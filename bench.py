@@ -186,17 +186,25 @@ def main():
         if has_gpu:
             torch.cuda.synchronize()
 
-    # hipGraph capture of the step: one replay per step instead of two eager
-    # launches — removes per-launch CPU overhead, which dominates the small
-    # end of the sweep (a 1 MiB step is ~13 us eager, most of it launch).
-    # RCCL collectives are capturable (comm bound via device_id above);
-    # falls back to eager if capture fails.
+    # hipGraph capture of the step (nccl-tests -G semantics: capture
+    # `iters_per_graph` iterations per graph) — one replay enqueues the
+    # whole captured loop, removing per-launch CPU overhead, which
+    # dominates the small end of the sweep (a 1 MiB step is ~13 us eager
+    # from python, most of it launch).  RCCL collectives are capturable
+    # (comm bound via device_id above); falls back to eager if capture
+    # fails.
     use_graphs = has_gpu and os.environ.get("CEA_BENCH_GRAPH", "1") == "1"
 
-    def make_step(buf_send, buf_recv):
-        """Returns (callable, is_graph) executing one timed step."""
+    def make_step(buf_send, buf_recv, iters_per_graph=1):
+        """Returns (callable, is_graph); the callable executes
+        `iters_per_graph` timed steps."""
         if not use_graphs:
-            return (lambda: step(buf_send, buf_recv)), False
+
+            def eager():
+                for _ in range(iters_per_graph):
+                    step(buf_send, buf_recv)
+
+            return eager, False
         try:
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
@@ -207,11 +215,17 @@ def main():
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                step(buf_send, buf_recv)
+                for _ in range(iters_per_graph):
+                    step(buf_send, buf_recv)
             return g.replay, True
         except Exception:  # noqa: BLE001 - capture support is optional
             torch.cuda.synchronize()
-            return (lambda: step(buf_send, buf_recv)), False
+
+            def eager():
+                for _ in range(iters_per_graph):
+                    step(buf_send, buf_recv)
+
+            return eager, False
 
     def busbw_factor(nranks: int) -> float:
         return 2.0 * (nranks - 1) / nranks if nranks > 1 else 1.0
@@ -223,14 +237,18 @@ def main():
         while size <= args.max_bytes:
             e = size // 4
             s_send, s_recv = send[:e], recv[:e]
-            run_step, _ = make_step(s_send, s_recv)
-            for _ in range(args.sweep_warmup):
-                run_step()
+            # whole sweep loop in one graph (-G sweep_iters equivalent):
+            # one replay = all iterations, so python launch cost is paid
+            # once per size instead of once per iteration.
+            run_iters, _ = make_step(s_send, s_recv,
+                                     iters_per_graph=args.sweep_iters)
+            run_warm, _ = make_step(s_send, s_recv,
+                                    iters_per_graph=args.sweep_warmup)
+            run_warm()
             sync()
             dist.barrier()
             t0 = time.perf_counter()
-            for _ in range(args.sweep_iters):
-                run_step()
+            run_iters()
             sync()
             el = time.perf_counter() - t0
             dist.barrier()
