@@ -34,11 +34,33 @@ def parse_version(version: str) -> Dict[str, str]:
     return out
 
 
+def rccl_version() -> str:
+    """RCCL release of the node userspace, via ncclGetVersion (version code
+    major*10000 + minor*100 + patch).  Empty string when librccl is absent
+    (CPU-only control-plane nodes)."""
+    import ctypes
+
+    for name in ("librccl.so", "librccl.so.1"):
+        try:
+            lib = ctypes.CDLL(name)
+            v = ctypes.c_int()
+            if lib.ncclGetVersion(ctypes.byref(v)) == 0 and v.value > 0:
+                code = v.value
+                return f"{code // 10000}.{code % 10000 // 100}.{code % 100}"
+        except OSError:
+            continue
+    return ""
+
+
 def publish_driver_version_annotations(kube_client, node_name: str) -> None:
-    """Parity: PublishDriverVersionAnnotations (version_visibility.go:38-46)."""
+    """Parity: PublishDriverVersionAnnotations (version_visibility.go:38-46);
+    extended with the RCCL release (the transport the L4 layer enables)."""
     ops = amdsmi.get_ops()
     version = ops.driver_version()
     annotations = parse_version(version)
     annotations["amd.com/rocm.release"] = ops.rocm_version()
+    rccl = rccl_version()
+    if rccl:
+        annotations["amd.com/rccl.version"] = rccl
     kube_client.apply_node_annotations(node_name, annotations, FIELD_MANAGER)
     log.info("published driver version annotations: %s", annotations)

@@ -189,6 +189,11 @@ def test_publish_annotations(tmp_path):
     ann = kube.nodes["n1"]["metadata"]["annotations"]
     assert ann["amd.com/rocm.driver-version.full"] == "6.10.5"
     assert ann["amd.com/rocm.release"] == "7.2.0"
+    # librccl ships in the ROCm image this runs in, so the RCCL release
+    # annotation must be present and dotted (e.g. "2.27.7")
+    from cea_amd.deviceplugin import version_visibility as vv
+    if vv.rccl_version():
+        assert ann["amd.com/rccl.version"].count(".") == 2
 
 
 # -- entrypoint config parsing (nvidia_gpu.go:64-108 parity) ------------------
