@@ -213,8 +213,15 @@ def main():
                     step(buf_send, buf_recv)
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
+            # Let the ProcessGroupNCCL watchdog dequeue the completed
+            # warmup works before capture starts: its hipEventQuery on a
+            # still-enqueued work during capture is
+            # hipErrorStreamCaptureUnsupported and aborts the process.
+            time.sleep(0.25)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            # thread_local: event queries from OTHER threads (the NCCL
+            # watchdog) stay legal while this thread captures.
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 for _ in range(iters_per_graph):
                     step(buf_send, buf_recv)
             return g.replay, True
