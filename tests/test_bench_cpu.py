@@ -62,6 +62,8 @@ def test_bench_two_process_gloo():
     assert result["ms_per_step"] > 0
     # busbw factor for n=2 is 2*(2-1)/2 = 1.0
     assert abs(result["config"]["busbw_factor"] - 1.0) < 1e-6
+    # n>=2 times the in-place collective (rccl-tests in-place row)
+    assert result["config"]["mode"] == "in-place"
     assert len(result["config"]["sweep"]) == 3  # 256K, 512K, 1M
     assert result["config"]["allocate_p50_us"] is None or result["config"]["allocate_p50_us"] > 0
     # rank 1 must not print the JSON line
