@@ -298,3 +298,43 @@ def test_hsa_cu_mask_fencing_throughput():
     assert t_fenced > 2.0 * t_full, (
         f"CU mask did not fence compute: full={t_full:.3f}s fenced={t_fenced:.3f}s"
     )
+
+
+def test_metrics_http_endpoint_on_hardware():
+    """End-to-end /metrics scrape with the real amdsmi backend: node-level
+    gauges must be served over HTTP with make/accelerator_id/model labels
+    (parity: MetricServer + promhttp, metrics.go:120-161)."""
+    import socket
+    import urllib.request
+
+    import cea_amd.amdsmi as amdsmi
+    from cea_amd.amdsmi.shim import ShimAmdSmi
+    from cea_amd.deviceplugin.manager import AmdGPUManager, GPUConfig
+    from cea_amd.deviceplugin.metrics import MetricServer
+
+    smi = ShimAmdSmi()
+    smi.init()
+    amdsmi.set_ops(smi)
+    try:
+        mgr = AmdGPUManager(GPUConfig())
+        mgr.config.add_defaults_and_validate()
+        mgr.discover_gpus()
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        srv = MetricServer(mgr, port=port, collection_interval_s=3600)
+        srv.start()
+        try:
+            srv.collect_once()  # pod-resources absent on the box: node-level only
+            body = urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/metrics", timeout=10
+            ).read().decode()
+            assert 'duty_cycle_gpu_node{' in body
+            assert 'memory_total_gpu_node{' in body
+            assert 'make="amd"' in body
+        finally:
+            srv.stop()
+    finally:
+        smi.shutdown()
+        amdsmi.ops = None
