@@ -16,7 +16,6 @@ restart watch, manager.go:442-549) is kept trigger-for-trigger.
 from __future__ import annotations
 
 import dataclasses
-import glob
 import logging
 import os
 import queue

@@ -22,7 +22,7 @@ import time
 from typing import Dict, List, Optional
 
 import grpc
-from prometheus_client import Gauge, start_http_server, REGISTRY
+from prometheus_client import Gauge, start_http_server
 
 from .. import amdsmi
 from ..kube import protos as api

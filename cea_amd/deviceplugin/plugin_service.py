@@ -13,7 +13,6 @@ from __future__ import annotations
 
 import logging
 import queue
-import time
 
 import grpc
 

@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import ctypes
 import os
-from typing import Optional, Tuple
+from typing import Optional
 
 _LIB_NAME = "libceaamd_gpu.so"
 _lib: Optional[ctypes.CDLL] = None

@@ -23,7 +23,6 @@ import time
 from collections import defaultdict
 from typing import Dict, List, Optional
 
-from ..deviceplugin import RESOURCE_NAME
 from .topology import CandidateNode, assign_pods, topo_key_from_labels
 
 log = logging.getLogger(__name__)
