@@ -63,9 +63,20 @@ class KubeClient:
              content_type: str = "application/json"):
         url = self.base_url + path
         headers = {"Content-Type": content_type}
+        json_body = None
+        data = None
+        if body is not None:
+            if content_type.endswith("json"):
+                json_body = body
+            else:
+                # apply-patch+yaml (server-side apply): JSON is a YAML
+                # subset, so a JSON-serialized body is valid — passing the
+                # dict as `data` would form-encode it.
+                import json as _json
+
+                data = _json.dumps(body)
         r = self.session.request(
-            method, url, json=body if content_type.endswith("json") else None,
-            data=None if content_type.endswith("json") else body,
+            method, url, json=json_body, data=data,
             params=params, headers=headers, timeout=30,
         )
         if r.status_code >= 400:
