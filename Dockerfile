@@ -17,7 +17,7 @@ COPY cea_amd/ cea_amd/
 COPY cmd/ cmd/
 COPY bench.py ./
 
-RUN make all
+RUN make all && ln -s /opt/cea-amd/cea_amd/bin /opt/cea-amd/bin
 
 ENV PYTHONPATH=/opt/cea-amd
 ENTRYPOINT ["python3", "/opt/cea-amd/cmd/amd_gpu.py"]

@@ -67,3 +67,22 @@ def test_rccl_config_has_sweep_protocol():
         assert frag in script, frag
     env = cm["data"]["rccl-env.sh"]
     assert "HSA_ENABLE_IPC_MODE_LEGACY=0" in env
+
+
+def test_image_paths_referenced_by_manifests_exist_in_tree():
+    """Every /opt/cea-amd/<path> a manifest invokes must exist in the repo
+    (the Dockerfile copies the tree to /opt/cea-amd and symlinks bin/)."""
+    import re
+    docs = load_all()
+    pat = re.compile(r"/opt/cea-amd/([\w./-]+)")
+    for path, _ in docs.items():
+        with open(path) as f:
+            for ref in pat.findall(f.read()):
+                local = ref
+                if local.startswith("bin/"):  # Dockerfile: bin -> cea_amd/bin
+                    local = "cea_amd/" + local
+                # binaries are built by `make all`; map to their sources
+                built = {"cea_amd/bin/all_reduce_perf": "csrc/all_reduce_perf.cpp"}
+                local = built.get(local, local)
+                assert os.path.exists(os.path.join(REPO, local)), \
+                    f"{path} references /opt/cea-amd/{ref} not present in tree"
