@@ -86,3 +86,37 @@ def test_image_paths_referenced_by_manifests_exist_in_tree():
                 local = built.get(local, local)
                 assert os.path.exists(os.path.join(REPO, local)), \
                     f"{path} references /opt/cea-amd/{ref} not present in tree"
+
+
+def test_configmap_references_resolve():
+    """Every configMap a manifest mounts or envFrom's must be defined by
+    some manifest in the tree (catches rename drift between installers and
+    test pods).  Secrets are allowed to be cluster-created (the notebook
+    README documents its token secret)."""
+    defined = set()
+    referenced = set()
+
+    def walk(obj):
+        if isinstance(obj, dict):
+            if obj.get("kind") == "ConfigMap" and "metadata" in obj:
+                defined.add(obj["metadata"]["name"])
+            cm = obj.get("configMap")
+            if isinstance(cm, dict) and "name" in cm:
+                referenced.add(cm["name"])
+            ref = obj.get("configMapRef")
+            if isinstance(ref, dict) and "name" in ref:
+                referenced.add(ref["name"])
+            key_ref = obj.get("configMapKeyRef")
+            if isinstance(key_ref, dict) and "name" in key_ref:
+                referenced.add(key_ref["name"])
+            for v in obj.values():
+                walk(v)
+        elif isinstance(obj, list):
+            for v in obj:
+                walk(v)
+
+    for _, items in load_all().items():
+        for d in items:
+            walk(d)
+    missing = referenced - defined
+    assert not missing, f"configMaps referenced but never defined: {missing}"
