@@ -22,7 +22,10 @@ EVT_VMFAULT = 1            # AMDSMI_EVT_NOTIF_VMFAULT — GPU page fault
 EVT_THERMAL_THROTTLE = 2   # AMDSMI_EVT_NOTIF_THERMAL_THROTTLE
 EVT_GPU_PRE_RESET = 3      # AMDSMI_EVT_NOTIF_GPU_PRE_RESET
 EVT_GPU_POST_RESET = 4     # AMDSMI_EVT_NOTIF_GPU_POST_RESET
-EVT_RING_HANG = 5          # AMDSMI_EVT_NOTIF_RING_HANG
+EVT_PAGE_FAULT_START = 7   # AMDSMI_EVT_NOTIF_PAGE_FAULT_START
+# NOTE: this ROCm's amdsmi enum has no RING_HANG; 5/6 are MIGRATE_START/END
+# (benign memory-migration traffic) and are deliberately neither armed nor
+# monitored.
 # Synthetic codes (outside the amdsmi notification enum) raised by the
 # polling side of the health checker:
 EVT_ECC_UNCORRECTABLE = 48  # uncorrectable ECC count increased (parity with
@@ -37,7 +40,7 @@ DEFAULT_HEALTH_CRITICAL_EVENTS = {EVT_ECC_UNCORRECTABLE}
 MONITOR_CRITICAL_EVENTS = {
     EVT_VMFAULT,
     EVT_GPU_PRE_RESET,
-    EVT_RING_HANG,
+    EVT_PAGE_FAULT_START,
     EVT_ECC_UNCORRECTABLE,
     EVT_XGMI_ERROR,
     EVT_LOST,
@@ -125,7 +128,7 @@ class AmdSmiOperations(ABC):
     @abstractmethod
     def wait_events(self, timeout_ms: int) -> List[Event]:
         """Blocking dequeue of device events (vm fault / reset / thermal /
-        ring hang), parity with nvml.WaitForEvent (health_checker.go:461)."""
+        page fault), parity with nvml.WaitForEvent (health_checker.go:461)."""
 
 
 def numa_node_for_bdf(bdf: str, sysfs_root: str = "/sys") -> Optional[int]:

@@ -3,7 +3,7 @@
 Role parity: /root/reference/pkg/gpu/nvidia/health_check/health_checker.go
 (476 LoC), re-based on AMD-SMI:
   * NVML Xid events -> amdsmi event notifications (VM fault, thermal
-    throttle, GPU pre/post reset, ring hang) via the native shim's
+    throttle, GPU pre/post reset, page fault) via the native shim's
     wait_events, PLUS a polling watchdog over uncorrectable-ECC counters
     and device reachability (AMD-SMI has no ECC *event*, so code 48/79
     parity comes from polling deltas).
