@@ -102,8 +102,6 @@ def test_fault_marks_device_unhealthy_end_to_end(tmp_path):
             "try: torch.cuda.synchronize()\n"
             "except Exception: pass\n"
         )
-        subprocess.run([sys.executable, "-c", code], env=env, timeout=240,
-                       capture_output=True)
 
         # the health checker's 5s event wait must push an Unhealthy resend
         got_unhealthy = threading.Event()
@@ -119,7 +117,16 @@ def test_fault_marks_device_unhealthy_end_to_end(tmp_path):
 
         t = threading.Thread(target=reader, daemon=True)
         t.start()
-        assert got_unhealthy.wait(60), \
+        # KFD SMI event delivery to a freshly armed notification fd is
+        # occasionally dropped on the first fault (see the retry in
+        # test_vmfault_event_reaches_health_path); fire up to twice.
+        for attempt in range(2):
+            r = subprocess.run([sys.executable, "-c", code], env=env,
+                               timeout=240, capture_output=True, text=True)
+            assert r.returncode != 0, f"no GPU fault raised\n{r.stdout}{r.stderr}"
+            if got_unhealthy.wait(30):
+                break
+        assert got_unhealthy.is_set(), \
             "fault did not propagate to ListAndWatch as Unhealthy"
         stream.cancel()
         client.close()
