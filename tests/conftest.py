@@ -14,6 +14,17 @@ def pytest_configure(config):
     )
 
 
+def pytest_collection_modifyitems(config, items):
+    """Default per-test timeout on the GPU tier (pytest-timeout): a hung
+    kernel or event wait must fail the one test, not stall the whole
+    hardware run."""
+    if not config.pluginmanager.hasplugin("timeout"):
+        return
+    for item in items:
+        if item.get_closest_marker("gpu") and not item.get_closest_marker("timeout"):
+            item.add_marker(pytest.mark.timeout(600, method="thread"))
+
+
 @pytest.fixture(autouse=True)
 def _reset_globals():
     """Reset package-global injection points between tests (the same hygiene
