@@ -122,7 +122,8 @@ def vector_add(a, b, c, inject_fault: bool = False, stream=None) -> None:
 
 def copy_(dst, src, stream=None) -> None:
     """Async d2d copy dst <- src (contiguous CUDA tensors, same nbytes,
-    multiple of 16 bytes) with the float4-vectorized gfx950 kernel."""
+    multiple of 16 bytes) with the tuned gfx950 kernel (exact-cover 16-B
+    vector lanes; nontemporal past the 256 MiB Infinity Cache)."""
     import torch
 
     lib = load_library()
