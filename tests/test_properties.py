@@ -1,0 +1,92 @@
+"""Property-based tests (hypothesis) for the pure seams: virtual-ID
+round-trips, version parsing, topology scoring, partition-table parsing.
+These are the functions whose inputs come from external text (amd-smi output,
+kubelet requests, driver version strings) — fuzz them."""
+
+from hypothesis import given, settings, strategies as st
+
+from cea_amd.deviceplugin import sharing
+from cea_amd.deviceplugin.version_visibility import parse_version, ANNOTATION_PREFIX
+from cea_amd.partition.partition_gpu import parse_partition_status
+from cea_amd.scheduler import topology as topo
+
+
+dev_ids = st.one_of(
+    st.integers(0, 63).map(lambda i: f"amdgpu{i}"),
+    st.tuples(st.integers(0, 7), st.integers(0, 7)).map(
+        lambda t: f"amdgpu{t[0]}/xcd{t[1]}"
+    ),
+)
+
+
+@given(dev_ids, st.integers(0, 255))
+def test_virtual_id_round_trip(phys, idx):
+    v = sharing.virtual_id(phys, idx)
+    assert sharing.is_virtual_id(v)
+    assert not sharing.is_virtual_id(phys)
+    assert sharing.virtual_to_physical(v) == phys
+
+
+@given(st.text(max_size=40))
+def test_is_virtual_id_never_raises(s):
+    sharing.is_virtual_id(s)
+
+
+@given(st.text(alphabet=st.characters(blacklist_categories=("Cs",)), max_size=40))
+def test_parse_version_total(s):
+    """parse_version must never raise and always include 'full'."""
+    out = parse_version(s)
+    assert out[f"{ANNOTATION_PREFIX}.full"] == s
+    if f"{ANNOTATION_PREFIX}.major" in out:
+        assert out[f"{ANNOTATION_PREFIX}.major"].isdigit()
+
+
+@given(st.text(max_size=2000))
+def test_parse_partition_status_never_raises(s):
+    out = parse_partition_status(s)
+    for d in out:
+        assert "accelerator_partition" in d
+
+
+@given(
+    st.lists(
+        st.tuples(st.integers(0, 3), st.integers(0, 2), st.integers(0, 4)),
+        min_size=1,
+        max_size=10,
+    )
+)
+@settings(max_examples=50)
+def test_pairwise_score_matches_bruteforce_property(keys):
+    topos = [topo.TopoKey(f"b{a}", f"s{b}", f"h{c}") for a, b, c in keys]
+    brute = sum(
+        topo.distance(topos[i], topos[j])
+        for i in range(len(topos))
+        for j in range(i + 1, len(topos))
+    )
+    assert topo._pairwise_score(topos) == brute
+
+
+@given(
+    st.integers(1, 30),
+    st.lists(
+        st.tuples(st.integers(0, 5), st.integers(0, 4)), min_size=1, max_size=20
+    ),
+)
+@settings(max_examples=50)
+def test_assign_pods_fits_or_none(num_pods, node_spec):
+    nodes = [
+        topo.CandidateNode(f"n{i}", topo.TopoKey(f"b{b}", "s", f"h{i}"), cap)
+        for i, (b, cap) in enumerate(node_spec)
+    ]
+    got = topo.assign_pods(num_pods, nodes)
+    total_cap = sum(c for _, c in node_spec)
+    if total_cap < num_pods:
+        assert got is None
+    else:
+        assert got is not None and len(got) == num_pods
+        # no node exceeds its capacity
+        from collections import Counter
+        counts = Counter(got)
+        caps = {f"n{i}": c for i, (_, c) in enumerate(node_spec)}
+        for name, used in counts.items():
+            assert used <= caps[name], (name, used)
