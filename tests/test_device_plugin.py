@@ -205,3 +205,57 @@ def test_hot_add_restart_trigger(tmp_path):
     assert not mgr.has_additional_gpus_installed()
     open(os.path.join(mgr.dev_directory, "dri", "renderD200"), "w").close()
     assert mgr.has_additional_gpus_installed()
+
+
+def test_concurrent_allocate_and_health_churn(tmp_path):
+    """Stress the serve path: parallel Allocate callers while the health
+    channel churns resends on the ListAndWatch stream — no deadlock, no
+    error, stream keeps flowing (guards the grpc thread-pool sizing:
+    streams pin worker threads)."""
+    import queue as _queue
+
+    from cea_amd.kube import protos as api
+
+    mgr, _ = make_manager(tmp_path, num_gpus=4)
+    serve_in_thread(mgr)
+    sock = os.path.join(mgr.plugin_directory, mgr.socket_name)
+    client = PluginClient(sock)
+    stream = client.list_and_watch_once()
+    it = iter(stream)
+    next(it)  # initial send
+
+    errors: "_queue.Queue[Exception]" = _queue.Queue()
+
+    def churn_health():
+        for i in range(20):
+            mgr.health.put(api.Device(
+                ID=f"amdgpu{i % 4}",
+                health=api.UNHEALTHY if i % 2 else api.HEALTHY))
+            time.sleep(0.01)
+
+    def allocate_loop():
+        c = PluginClient(sock)
+        try:
+            for _ in range(25):
+                resp = c.allocate([["amdgpu1"]])
+                assert resp.container_responses[0].devices
+        except Exception as e:  # noqa: BLE001
+            errors.put(e)
+        finally:
+            c.close()
+
+    threads = [threading.Thread(target=churn_health)] + [
+        threading.Thread(target=allocate_loop) for _ in range(4)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+        assert not t.is_alive(), "worker hung"
+    assert errors.empty(), errors.get()
+    # the stream must still deliver resends after the churn
+    resp = next(it)
+    assert len(resp.devices) == 4
+    stream.cancel()
+    client.close()
+    mgr.stop()
