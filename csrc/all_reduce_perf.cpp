@@ -57,7 +57,24 @@ struct Args {
   int graph = 0;                   // -G: capture the op in a hipGraph and
                                    //     time graph replays (nccl-tests -G)
   std::string op = "all_reduce";   // -o
+  std::string dtype = "float";     // -d (nccl-tests datatype flag)
 };
+
+struct DType {
+  ncclDataType_t nccl;
+  int size;
+};
+
+DType parse_dtype(const std::string& s) {
+  if (s == "float" || s == "fp32") return {ncclFloat, 4};
+  if (s == "half" || s == "fp16") return {ncclHalf, 2};
+  if (s == "bfloat16" || s == "bf16") return {ncclBfloat16, 2};
+  if (s == "double" || s == "fp64") return {ncclDouble, 8};
+  if (s == "fp8e4m3") return {ncclFloat8e4m3, 1};
+  if (s == "fp8e5m2") return {ncclFloat8e5m2, 1};
+  fprintf(stderr, "unknown dtype %s\n", s.c_str());
+  exit(1);
+}
 
 long parse_size(const char* s) {
   char* end = nullptr;
@@ -83,6 +100,7 @@ double busbw_factor(const std::string& op, int n) {
 
 struct Ctx {
   int n;
+  DType dt{ncclFloat, 4};
   std::vector<ncclComm_t> comms;
   std::vector<hipStream_t> streams;
   std::vector<float*> send;
@@ -94,16 +112,16 @@ void run_op(Ctx& c, const std::string& op, long count) {
   NCCLCHECK(ncclGroupStart());
   for (int i = 0; i < c.n; i++) {
     if (op == "all_reduce") {
-      NCCLCHECK(ncclAllReduce(c.send[i], c.recv[i], count, ncclFloat, ncclSum,
+      NCCLCHECK(ncclAllReduce(c.send[i], c.recv[i], count, c.dt.nccl, ncclSum,
                               c.comms[i], c.streams[i]));
     } else if (op == "all_gather") {
-      NCCLCHECK(ncclAllGather(c.send[i], c.recv[i], count / c.n, ncclFloat,
+      NCCLCHECK(ncclAllGather(c.send[i], c.recv[i], count / c.n, c.dt.nccl,
                               c.comms[i], c.streams[i]));
     } else if (op == "reduce_scatter") {
-      NCCLCHECK(ncclReduceScatter(c.send[i], c.recv[i], count / c.n, ncclFloat,
+      NCCLCHECK(ncclReduceScatter(c.send[i], c.recv[i], count / c.n, c.dt.nccl,
                                   ncclSum, c.comms[i], c.streams[i]));
     } else if (op == "broadcast") {
-      NCCLCHECK(ncclBroadcast(c.send[i], c.recv[i], count, ncclFloat, 0,
+      NCCLCHECK(ncclBroadcast(c.send[i], c.recv[i], count, c.dt.nccl, 0,
                               c.comms[i], c.streams[i]));
     } else {
       fprintf(stderr, "unknown op %s\n", op.c_str());
@@ -138,10 +156,11 @@ int main(int argc, char** argv) {
     else if (f == "-n" || f == "--iters") a.iters = atoi(next());
     else if (f == "-c") a.check = atoi(next());
     else if (f == "-G") a.graph = atoi(next());
+    else if (f == "-d") a.dtype = next();
     else if (f == "-o") a.op = next();
     else if (f == "-h" || f == "--help") {
       printf("usage: %s [-b min] [-e max] [-f factor] [-g ngpus] [-w warmup]"
-             " [-n iters] [-c check] [-G graph] [-o all_reduce|all_gather|"
+             " [-n iters] [-c check] [-G graph] [-d dtype] [-o all_reduce|all_gather|"
              "reduce_scatter|broadcast]\n", argv[0]);
       return 0;
     }
@@ -157,33 +176,39 @@ int main(int argc, char** argv) {
 
   Ctx c;
   c.n = a.ngpus;
+  c.dt = parse_dtype(a.dtype);
+  if (a.check && a.dtype != "float") {
+    fprintf(stderr, "-c 1 only supported with -d float; disabling check\n");
+    a.check = 0;
+  }
   c.comms.resize(c.n);
   c.streams.resize(c.n);
   c.send.resize(c.n);
   c.recv.resize(c.n);
-  long max_count = a.max_bytes / 4;
+  long max_count = a.max_bytes / c.dt.size;
   for (int i = 0; i < c.n; i++) {
     HIPCHECK(hipSetDevice(i));
     HIPCHECK(hipStreamCreate(&c.streams[i]));
     HIPCHECK(hipMalloc(&c.send[i], a.max_bytes));
     HIPCHECK(hipMalloc(&c.recv[i], a.max_bytes));
-    hipLaunchKernelGGL(fill_kernel, dim3(1024), dim3(256), 0, 0, c.send[i],
-                       max_count, (float)(i + 1));
+    hipLaunchKernelGGL(fill_kernel, dim3(1024), dim3(256), 0, 0,
+                       (float*)c.send[i], a.max_bytes / 4, (float)(i + 1));
     HIPCHECK(hipDeviceSynchronize());
   }
   NCCLCHECK(ncclCommInitAll(c.comms.data(), c.n, nullptr));
 
   int rccl_major = 0, rccl_minor = 0, rccl_patch = 0;
   ncclGetVersion(&rccl_major);
-  printf("# all_reduce_perf-equivalent (cea_amd, RCCL %d) op=%s nGpus=%d "
-         "warmup=%d iters=%d check=%d\n",
-         rccl_major, a.op.c_str(), c.n, a.warmup, a.iters, a.check);
+  printf("# all_reduce_perf-equivalent (cea_amd, RCCL %d) op=%s dtype=%s "
+         "nGpus=%d warmup=%d iters=%d check=%d\n",
+         rccl_major, a.op.c_str(), a.dtype.c_str(), c.n, a.warmup, a.iters,
+         a.check);
   printf("#%12s %12s %10s %10s %10s\n", "size(B)", "count", "time(us)",
          "algbw(GB/s)", "busbw(GB/s)");
   (void)rccl_minor; (void)rccl_patch;
 
   for (long bytes = a.min_bytes; bytes <= a.max_bytes; bytes *= a.factor) {
-    long count = bytes / 4;
+    long count = bytes / c.dt.size;
 
     // -G: capture one grouped op into a hipGraph per stream, then time
     // whole-graph replays (one replay = one collective) — removes the
