@@ -172,7 +172,19 @@ def main(argv=None):
 
         threading.Thread(target=_publish, daemon=True).start()
 
-    mgr.serve()   # blocks forever (parity nvidia_gpu.go:225)
+    # Graceful pod termination: kubelet sends SIGTERM; stop the serve loop
+    # so the gRPC server closes and the plugin socket is released cleanly
+    # (kubelet re-registers us on restart either way).
+    import signal
+
+    def _terminate(signum, frame):  # noqa: ARG001
+        log.info("received signal %d; shutting down", signum)
+        mgr.stop()
+
+    signal.signal(signal.SIGTERM, _terminate)
+    signal.signal(signal.SIGINT, _terminate)
+
+    mgr.serve()   # blocks until stopped (parity nvidia_gpu.go:225)
 
 
 if __name__ == "__main__":

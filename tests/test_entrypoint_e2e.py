@@ -81,3 +81,18 @@ def test_entrypoint_time_sharing_config(tmp_path):
         client.close()
     finally:
         stop(proc, stub)
+
+
+def test_entrypoint_graceful_sigterm(tmp_path):
+    """SIGTERM (kubelet pod termination) must exit 0 within the watchdog
+    poll interval, with the serve loop shut down cleanly."""
+    import signal
+
+    proc, stub, plugin_dir = run_entrypoint(tmp_path, num_gpus=1)
+    try:
+        assert stub.registered.wait(30), "entrypoint never registered"
+        proc.send_signal(signal.SIGTERM)
+        rc = proc.wait(timeout=15)
+        assert rc == 0, proc.stdout.read()[-1000:]
+    finally:
+        stop(proc, stub)
