@@ -195,19 +195,31 @@ class GPUHealthChecker:
     def _affected_device_ids(self, ev: Event):
         """No UUID => every device (parity health_checker.go:415-424); with a
         UUID => all partitions of the faulting die (the CPX analog of MIG
-        UUID+GI/CI matching, health_checker.go:426-445)."""
+        UUID+GI/CI matching, health_checker.go:426-445).  A UUID that
+        matches NO enumerated device also marks every device: a critical
+        event on this node must never be silently dropped because the
+        event source formats the id differently than enumeration."""
         pm = self.manager.partition_manager
+        all_ids = (list(pm.devices.keys()) if pm
+                   else list(self.manager.devices.keys()))
         if not ev.device_uuid:
-            if pm:
-                return list(pm.devices.keys())
-            return list(self.manager.devices.keys())
+            return all_ids
+        uuid = ev.device_uuid.strip().lower()
         if pm:
-            return pm.devices_for_die_uuid(ev.device_uuid)
-        return [
-            dev_id
-            for dev_id, info in self.manager.device_infos.items()
-            if info.uuid == ev.device_uuid
-        ]
+            matched = pm.devices_for_die_uuid(ev.device_uuid)
+        else:
+            matched = [
+                dev_id
+                for dev_id, info in self.manager.device_infos.items()
+                if info.uuid.strip().lower() == uuid
+            ]
+        if not matched:
+            log.warning(
+                "event uuid %s matches no enumerated device; marking all",
+                ev.device_uuid,
+            )
+            return all_ids
+        return matched
 
     # -- kube plumbing ---------------------------------------------------------
     def _record_event(self, ev: Event) -> None:

@@ -212,3 +212,14 @@ def test_xgmi_error_poll_raises_code_63(tmp_path):
         assert len(kube.events) == n_events
     finally:
         hc.stop()
+
+
+def test_critical_event_with_unknown_uuid_marks_all(tmp_path):
+    """A critical event whose UUID matches no enumerated device marks every
+    device: never silently drop a node-level critical signal over an id
+    formatting mismatch between the event source and enumeration."""
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc.catch_error(Event(device_uuid="GPU-UNKNOWN-FORMAT",
+                         code=EVT_ECC_UNCORRECTABLE))
+    ids = sorted(d.ID for d in drain(mgr.health))
+    assert ids == ["amdgpu0", "amdgpu1"]
