@@ -77,12 +77,13 @@ def test_fault_marks_device_unhealthy_end_to_end(tmp_path):
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     env["EVENT_CONFIG"] = "1,7"   # VMFAULT / page-fault => health-critical
     env.pop("NODE_NAME", None)    # no kube client in this tier
+    plugin_log = open(str(tmp_path / "plugin.log"), "w+")
     proc = subprocess.Popen(
         [sys.executable, os.path.join(REPO, "cmd", "amd_gpu.py"),
          "--plugin-directory", plugin_dir,
          "--enable-health-monitoring",
          "--gpu-config", str(tmp_path / "missing.json")],
-        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        env=env, stdout=plugin_log, stderr=subprocess.STDOUT, text=True,
     )
     try:
         assert stub.registered.wait(60)
@@ -126,8 +127,14 @@ def test_fault_marks_device_unhealthy_end_to_end(tmp_path):
             assert r.returncode != 0, f"no GPU fault raised\n{r.stdout}{r.stderr}"
             if got_unhealthy.wait(30):
                 break
-        assert got_unhealthy.is_set(), \
-            "fault did not propagate to ListAndWatch as Unhealthy"
+        if not got_unhealthy.is_set():
+            plugin_log.flush()
+            plugin_log.seek(0)
+            tail = plugin_log.read()[-3000:]
+            raise AssertionError(
+                "fault did not propagate to ListAndWatch as Unhealthy; "
+                f"plugin log tail:\n{tail}"
+            )
         stream.cancel()
         client.close()
     finally:
