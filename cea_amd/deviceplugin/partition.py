@@ -116,4 +116,9 @@ class PartitionDeviceManager:
         return self.device_specs.get(dev_id)
 
     def devices_for_die_uuid(self, uuid: str) -> List[str]:
-        return self.die_to_devices.get(uuid, [])
+        # normalized lookup, same policy as the health checker's matching
+        want = uuid.strip().lower()
+        for k, ids in self.die_to_devices.items():
+            if k.strip().lower() == want:
+                return ids
+        return []
