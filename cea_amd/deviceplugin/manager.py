@@ -116,6 +116,7 @@ class AmdGPUManager:
         container_path: str = "/usr/local/amd",
         plugin_directory: str = "/device-plugin",
         sysfs_root: str = "/sys",
+        extra_mounts: Optional[List[dict]] = None,
     ):
         self.config = config
         self.dev_directory = dev_directory
@@ -123,6 +124,10 @@ class AmdGPUManager:
         self.container_path = container_path
         self.plugin_directory = plugin_directory
         self.sysfs_root = sysfs_root
+        # optional additional read-only mounts (e.g. the OpenCL vendor ICD —
+        # the MI355X-compute analog of the reference's Vulkan ICD mounts,
+        # nvidia_gpu.go:50-61); only mounted when the host path exists
+        self.extra_mounts = list(extra_mounts or [])
 
         self.devices: Dict[str, api.Device] = {}
         self.device_infos: Dict[str, "amdsmi.DeviceInfo"] = {}
@@ -248,14 +253,23 @@ class AmdGPUManager:
 
     def mounts(self) -> List[dict]:
         """Driver/userspace tree mount, parity with the reference mounting
-        /home/kubernetes/bin/nvidia (nvidia_gpu.go:113-115)."""
-        return [
+        /home/kubernetes/bin/nvidia (nvidia_gpu.go:113-115), plus any
+        existing extra mounts (OpenCL ICD analog of the Vulkan ICD paths)."""
+        out = [
             {
                 "host_path": self.host_path,
                 "container_path": self.container_path,
                 "read_only": True,
             }
         ]
+        for m in self.extra_mounts:
+            if os.path.exists(m["host_path"]):
+                out.append({
+                    "host_path": m["host_path"],
+                    "container_path": m["container_path"],
+                    "read_only": True,
+                })
+        return out
 
     def envs(self, num_virtual_requested: int) -> Dict[str, str]:
         """Per-container env fencing for the cu-fencing strategy: the

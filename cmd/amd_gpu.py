@@ -35,6 +35,11 @@ def parse_args(argv=None):
     p.add_argument("--gpu-config", default="/etc/amd/gpu_config.json")
     p.add_argument("--gpu-fraction-divisor-file",
                    default="/etc/amd/gpu-fraction-divisor.txt")
+    p.add_argument("--opencl-icd-host-path", default="/etc/OpenCL/vendors",
+                   help="host dir with the OpenCL vendor ICD (the Vulkan-ICD"
+                        " analog, nvidia_gpu.go:50-61); mounted read-only "
+                        "into GPU containers when present")
+    p.add_argument("--opencl-icd-container-path", default="/etc/OpenCL/vendors")
     p.add_argument("--publish-driver-version", action="store_true")
     p.add_argument("--mock-amdsmi", action="store_true",
                    help="use the mock AMD-SMI backend (CPU e2e testing; the "
@@ -98,6 +103,10 @@ def main(argv=None):
         host_path=args.host_path,
         container_path=args.container_path,
         plugin_directory=args.plugin_directory,
+        extra_mounts=[{
+            "host_path": args.opencl_icd_host_path,
+            "container_path": args.opencl_icd_container_path,
+        }],
     )
 
     # device-path retry loop (parity nvidia_gpu.go:146-154): wait for L0

@@ -264,3 +264,25 @@ def test_gpu_readiness_helpers(tmp_path):
     f = tmp_path / "ctype.txt"
     f.write_text("SEV-SNP\n")
     assert mod.confidential_node_type(str(f)) == "sev-snp"
+
+
+def test_extra_mounts_only_when_host_path_exists(tmp_path):
+    """OpenCL-ICD mount (the Vulkan-ICD analog): included read-only when the
+    host dir exists, omitted otherwise."""
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, 1)
+    mock = MockAmdSmi(dev)
+    mock.init()
+    amdsmi.set_ops(mock)
+    icd = tmp_path / "OpenCL" / "vendors"
+    mgr = AmdGPUManager(
+        GPUConfig(), dev_directory=dev, sysfs_root=str(tmp_path / "sys"),
+        extra_mounts=[{"host_path": str(icd),
+                       "container_path": "/etc/OpenCL/vendors"}])
+    mgr.start()
+    assert len(mgr.mounts()) == 1  # host dir absent
+    icd.mkdir(parents=True)
+    mounts = mgr.mounts()
+    assert len(mounts) == 2
+    assert mounts[1]["container_path"] == "/etc/OpenCL/vendors"
+    assert mounts[1]["read_only"] is True
