@@ -384,15 +384,68 @@ class AmdGPUManager:
                 request_serializer=lambda m: m.SerializeToString(),
                 response_deserializer=api.Empty.FromString,
             )
-            register(
-                api.RegisterRequest(
-                    version=api.DEVICE_PLUGIN_VERSION,
-                    endpoint=self.socket_name,
-                    resource_name=RESOURCE_NAME,
-                ),
-                timeout=10,
+            req = api.RegisterRequest(
+                version=api.DEVICE_PLUGIN_VERSION,
+                endpoint=self.socket_name,
+                resource_name=RESOURCE_NAME,
             )
+            req.options.get_preferred_allocation_available = True
+            register(req, timeout=10)
         log.info("registered %s with kubelet", RESOURCE_NAME)
+
+    def preferred_allocation(self, available, must_include, size: int):
+        """Die/NUMA-aware device selection for GetPreferredAllocation —
+        the reference stubs this RPC (beta_plugin.go:95-103); on MI355X it
+        is worth implementing: CPX partitions of one die share L3/HBM and
+        the xGMI endpoint, so packing a request onto as few dies as
+        possible keeps the workload local and leaves whole dies free for
+        the next gang member.  Greedy: honor must_include, then repeatedly
+        take candidates from the die with the most free partitions
+        (preferring dies already used by this request)."""
+        chosen: List[str] = []
+        seen = set()
+        for d in must_include:
+            if d not in seen:
+                chosen.append(d)
+                seen.add(d)
+        chosen = chosen[:size]
+
+        def die_of(dev_id: str) -> str:
+            base = (sharing.virtual_to_physical(dev_id)
+                    if sharing.is_virtual_id(dev_id) else dev_id)
+            return base.split("/")[0]
+
+        remaining: Dict[str, List[str]] = {}
+        for d in available:
+            if d in seen:
+                continue
+            remaining.setdefault(die_of(d), []).append(d)
+        for ids in remaining.values():
+            ids.sort()
+        used_dies = {die_of(d) for d in chosen}
+        while len(chosen) < size and remaining:
+            need = size - len(chosen)
+            # dies this request already touches come first; then the
+            # smallest die that still covers the need (exact-fit packing,
+            # keeps big dies whole); if none covers it, the largest.
+            fits = [k for k in remaining if len(remaining[k]) >= need]
+            pool = fits or list(remaining)
+            die = min(
+                pool,
+                key=lambda k: (
+                    k not in used_dies,
+                    len(remaining[k]) if fits else -len(remaining[k]),
+                    k,
+                ),
+            )
+            ids = remaining.pop(die)
+            used_dies.add(die)
+            for d in ids:
+                if len(chosen) >= size:
+                    break
+                chosen.append(d)
+                seen.add(d)
+        return chosen
 
     def _status_check_loop(self, socket_path: str,
                            kubelet_watch: FileWatcher) -> None:

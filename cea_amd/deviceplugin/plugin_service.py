@@ -28,7 +28,11 @@ class PluginService:
 
     # -- RPC handlers --------------------------------------------------------
     def get_device_plugin_options(self, request, context):
-        return api.DevicePluginOptions()
+        opts = api.DevicePluginOptions()
+        # unlike the reference (error stub, beta_plugin.go:95-103) this
+        # plugin implements die-aware preferred allocation
+        opts.get_preferred_allocation_available = True
+        return opts
 
     def list_and_watch(self, request, context):
         """Initial device list, then a resend whenever a health event drains
@@ -96,9 +100,19 @@ class PluginService:
         return resp
 
     def get_preferred_allocation(self, request, context):
-        # parity: error stub beta_plugin.go:95-103
-        log.error("GetPreferredAllocation should not be called")
-        return api.PreferredAllocationResponse()
+        """Die/NUMA-aware selection (an MI355X improvement over the
+        reference's error stub, beta_plugin.go:95-103): pack each container
+        request onto as few physical dies as possible."""
+        resp = api.PreferredAllocationResponse()
+        for creq in request.container_requests:
+            chosen = self.manager.preferred_allocation(
+                list(creq.available_deviceIDs),
+                list(creq.must_include_deviceIDs),
+                int(creq.allocation_size),
+            )
+            cresp = resp.container_responses.add()
+            cresp.device_ids.extend(chosen)
+        return resp
 
     def pre_start_container(self, request, context):
         log.error("PreStartContainer should not be called")

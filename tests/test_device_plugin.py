@@ -259,3 +259,67 @@ def test_concurrent_allocate_and_health_churn(tmp_path):
     stream.cancel()
     client.close()
     mgr.stop()
+
+
+def test_preferred_allocation_packs_dies(tmp_path):
+    """GetPreferredAllocation (implemented here; the reference error-stubs
+    it) must pack a request onto as few dies as possible, honor
+    must_include, and exact-fit small dies before breaking big ones."""
+    from cea_amd.kube import protos as api
+
+    cfg = GPUConfig(compute_partition="cpx")
+    mgr, _ = make_manager(tmp_path, config=cfg, num_gpus=2,
+                          partitions_per_gpu=8, compute_partition="CPX")
+    serve_in_thread(mgr)
+    sock = os.path.join(mgr.plugin_directory, mgr.socket_name)
+
+    channel = grpc.insecure_channel(f"unix://{sock}")
+    options = channel.unary_unary(
+        api.DP_GET_OPTIONS,
+        request_serializer=lambda m: m.SerializeToString(),
+        response_deserializer=api.DevicePluginOptions.FromString,
+    )(api.Empty(), timeout=5)
+    assert options.get_preferred_allocation_available
+
+    preferred = channel.unary_unary(
+        api.DP_GET_PREFERRED_ALLOCATION,
+        request_serializer=lambda m: m.SerializeToString(),
+        response_deserializer=api.PreferredAllocationResponse.FromString,
+    )
+
+    # die0 has 3 free partitions, die1 has 8: a request for 3 should take
+    # die0 whole (exact fit) instead of breaking die1
+    req = api.PreferredAllocationRequest()
+    c = req.container_requests.add()
+    c.available_deviceIDs.extend(
+        [f"amdgpu0/xcd{i}" for i in (1, 4, 6)]
+        + [f"amdgpu1/xcd{i}" for i in range(8)])
+    c.allocation_size = 3
+    got = list(preferred(req, timeout=5).container_responses[0].device_ids)
+    assert got == ["amdgpu0/xcd1", "amdgpu0/xcd4", "amdgpu0/xcd6"]
+
+    # must_include on die1 pulls the rest from die1
+    req = api.PreferredAllocationRequest()
+    c = req.container_requests.add()
+    c.available_deviceIDs.extend(
+        [f"amdgpu0/xcd{i}" for i in (1, 4, 6)]
+        + [f"amdgpu1/xcd{i}" for i in range(8)])
+    c.must_include_deviceIDs.append("amdgpu1/xcd5")
+    c.allocation_size = 4
+    got = list(preferred(req, timeout=5).container_responses[0].device_ids)
+    assert got[0] == "amdgpu1/xcd5"
+    assert all(d.startswith("amdgpu1/") for d in got), got
+    assert len(got) == 4
+
+    # need spanning dies: take the big die whole, then the small one
+    req = api.PreferredAllocationRequest()
+    c = req.container_requests.add()
+    c.available_deviceIDs.extend(
+        [f"amdgpu0/xcd{i}" for i in (1, 4)]
+        + [f"amdgpu1/xcd{i}" for i in range(8)])
+    c.allocation_size = 9
+    got = list(preferred(req, timeout=5).container_responses[0].device_ids)
+    assert len(got) == 9
+    assert sum(d.startswith("amdgpu1/") for d in got) == 8
+    channel.close()
+    mgr.stop()
