@@ -90,3 +90,26 @@ def test_assign_pods_fits_or_none(num_pods, node_spec):
         caps = {f"n{i}": c for i, (_, c) in enumerate(node_spec)}
         for name, used in counts.items():
             assert used <= caps[name], (name, used)
+
+
+@given(
+    st.lists(st.tuples(st.integers(0, 3), st.integers(0, 7)),
+             min_size=0, max_size=24, unique=True),
+    st.integers(0, 2),
+    st.integers(0, 12),
+)
+@settings(max_examples=80)
+def test_preferred_allocation_invariants(avail_keys, n_must, size):
+    from cea_amd.deviceplugin.manager import AmdGPUManager, GPUConfig
+
+    mgr = AmdGPUManager(GPUConfig())
+    available = [f"amdgpu{d}/xcd{x}" for d, x in avail_keys]
+    must = available[:n_must]
+    got = mgr.preferred_allocation(available, must, size)
+    # no duplicates, bounded by size, drawn from the known ids
+    assert len(got) == len(set(got))
+    assert len(got) == min(size, len(available))
+    assert set(got) <= set(available)
+    # must_include honored up to size
+    for m in must[:size]:
+        assert m in got
