@@ -33,6 +33,7 @@ from typing import Dict, Optional, Set
 from .. import amdsmi
 from ..amdsmi.iface import (
     EVT_ECC_UNCORRECTABLE,
+    EVT_GPU_POST_RESET,
     EVT_LOST,
     EVT_XGMI_ERROR,
     MONITOR_CRITICAL_EVENTS,
@@ -70,6 +71,7 @@ class GPUHealthChecker:
         boot_id_path: str = BOOT_ID_PATH,
         ecc_poll_interval_s: float = ECC_POLL_INTERVAL_S,
         heartbeat_interval_s: float = HEARTBEAT_INTERVAL_S,
+        recover_on_reset: bool = True,
     ):
         self.manager = manager
         self.kube = kube_client
@@ -83,6 +85,7 @@ class GPUHealthChecker:
         self.boot_id_path = boot_id_path
         self.ecc_poll_interval_s = ecc_poll_interval_s
         self.heartbeat_interval_s = heartbeat_interval_s
+        self.recover_on_reset = recover_on_reset
         self._stop = threading.Event()
         self._threads = []
         self._ecc_baseline: Dict[int, int] = {}
@@ -178,9 +181,14 @@ class GPUHealthChecker:
 
     # -- the policy core ------------------------------------------------------
     def catch_error(self, ev: Event) -> None:
-        """Parity: catchError (health_checker.go:395-449)."""
+        """Parity: catchError (health_checker.go:395-449), plus post-reset
+        recovery the reference cannot do (NVML has no usable post-reset
+        signal; a device stays Unhealthy until plugin restart)."""
         log.warning("GPU event code=%d uuid=%s msg=%s", ev.code, ev.device_uuid,
                     ev.message)
+        if ev.code == EVT_GPU_POST_RESET and self.recover_on_reset:
+            self._recover_after_reset(ev)
+            return
         if ev.code not in self.monitor_critical:
             return
         self._record_event(ev)
@@ -191,6 +199,34 @@ class GPUHealthChecker:
         for dev_id in affected:
             log.warning("marking device %s Unhealthy (event %d)", dev_id, ev.code)
             self.manager.health.put(api.Device(ID=dev_id, health=api.UNHEALTHY))
+
+    def _recover_after_reset(self, ev: Event) -> None:
+        """GPU_POST_RESET: if the device answers a probe again, re-mark the
+        partitions the reset touched Healthy — kubelet restores capacity
+        without a plugin restart."""
+        ops = amdsmi.get_ops()
+        affected = self._affected_device_ids(ev)
+        for dev_id in affected:
+            if self.manager.device_health.get(dev_id) != api.UNHEALTHY:
+                continue
+            try:
+                # probe through the same seam discovery uses
+                idx = self._device_index(dev_id)
+                if idx is None:
+                    continue
+                ops.memory_info(idx)
+            except Exception as e:  # noqa: BLE001 - still broken, stay Unhealthy
+                log.warning("post-reset probe failed for %s: %s", dev_id, e)
+                continue
+            log.warning("device %s recovered after GPU reset; marking Healthy",
+                        dev_id)
+            self.manager.health.put(api.Device(ID=dev_id, health=api.HEALTHY))
+
+    def _device_index(self, dev_id: str):
+        pm = self.manager.partition_manager
+        infos = pm.devices if pm else self.manager.device_infos
+        info = infos.get(dev_id)
+        return info.index if info is not None else None
 
     def _affected_device_ids(self, ev: Event):
         """No UUID => every device (parity health_checker.go:415-424); with a

@@ -223,3 +223,25 @@ def test_critical_event_with_unknown_uuid_marks_all(tmp_path):
                          code=EVT_ECC_UNCORRECTABLE))
     ids = sorted(d.ID for d in drain(mgr.health))
     assert ids == ["amdgpu0", "amdgpu1"]
+
+
+def test_post_reset_recovers_unhealthy_device(tmp_path):
+    """GPU_POST_RESET re-marks a previously Unhealthy device Healthy when it
+    answers a probe again — capacity returns without a plugin restart
+    (an improvement over the reference, which stays Unhealthy forever)."""
+    from cea_amd.amdsmi.iface import EVT_GPU_POST_RESET
+
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_ECC_UNCORRECTABLE))
+    for d in drain(mgr.health):
+        mgr.set_device_health(d.ID, d.health)   # what ListAndWatch does
+    assert mgr.device_health["amdgpu1"] == "Unhealthy"
+
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    recovered = drain(mgr.health)
+    assert [d.ID for d in recovered] == ["amdgpu1"]
+    assert recovered[0].health == "Healthy"
+
+    # a healthy device's reset event pushes nothing
+    hc.catch_error(Event(device_uuid="mock-uuid-0", code=EVT_GPU_POST_RESET))
+    assert drain(mgr.health) == []
