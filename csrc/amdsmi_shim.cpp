@@ -96,6 +96,9 @@ typedef struct {
 const char* cea_smi_last_error() { return g_err; }
 
 int cea_smi_init() {
+  // serialize concurrent first-callers (ctypes callers may race init)
+  static std::mutex init_mutex;
+  std::lock_guard<std::mutex> init_lk(init_mutex);
   if (g_inited.load()) return 0;
   amdsmi_status_t st = amdsmi_init(AMDSMI_INIT_AMD_GPUS);
   if (st != AMDSMI_STATUS_SUCCESS) return fail("amdsmi_init", st);
