@@ -95,11 +95,12 @@ def to_nri_device(entry: dict) -> api.LinuxDevice:
         major=os.major(st.st_rdev),
         minor=os.minor(st.st_rdev),
     )
-    if entry.get("file_mode"):
+    # explicit None checks: uid/gid/file_mode 0 are valid values
+    if entry.get("file_mode") is not None:
         dev.file_mode.value = int(entry["file_mode"])
-    if entry.get("uid"):
+    if entry.get("uid") is not None:
         dev.uid.value = int(entry["uid"])
-    if entry.get("gid"):
+    if entry.get("gid") is not None:
         dev.gid.value = int(entry["gid"])
     return dev
 
