@@ -45,6 +45,11 @@ def parse_args():
     p.add_argument("--no-sweep", action="store_true",
                    help="skip the per-size sweep (headline size only)")
     p.add_argument("--backend", default="", help="nccl|gloo (default: auto)")
+    p.add_argument("--op", default="all_reduce",
+                   choices=["all_reduce", "all_gather", "reduce_scatter",
+                            "broadcast"],
+                   help="collective to benchmark (nccl-tests -o parity; "
+                        "the driver contract uses the default all_reduce)")
     return p.parse_args()
 
 
@@ -179,8 +184,25 @@ def main():
 
             native.assert_native_available()
 
-        def step(buf_send, buf_recv):
-            dist.all_reduce(buf_recv)
+        if args.op == "all_reduce":
+
+            def step(buf_send, buf_recv):
+                dist.all_reduce(buf_recv)
+        elif args.op == "all_gather":
+            # rccl-tests semantics: per-rank input = size/n, output = size
+            def step(buf_send, buf_recv):
+                e = buf_recv.numel() // n * n
+                dist.all_gather_into_tensor(buf_recv[:e],
+                                            buf_send[: e // n])
+        elif args.op == "reduce_scatter":
+
+            def step(buf_send, buf_recv):
+                e = buf_send.numel() // n * n
+                dist.reduce_scatter_tensor(buf_recv[: e // n], buf_send[:e])
+        else:  # broadcast
+
+            def step(buf_send, buf_recv):
+                dist.broadcast(buf_recv, src=0)
 
     def sync():
         if has_gpu:
@@ -235,7 +257,14 @@ def main():
             return eager, False
 
     def busbw_factor(nranks: int) -> float:
-        return 2.0 * (nranks - 1) / nranks if nranks > 1 else 1.0
+        """nccl-tests conventions per collective."""
+        if nranks <= 1:
+            return 1.0
+        if args.op == "all_reduce":
+            return 2.0 * (nranks - 1) / nranks
+        if args.op in ("all_gather", "reduce_scatter"):
+            return (nranks - 1) / nranks
+        return 1.0  # broadcast
 
     # ---- per-size sweep (reference protocol: -w 5 --iters 100 -c 0) ------
     sweep = []
@@ -300,7 +329,7 @@ def main():
         except Exception as e:  # noqa: BLE001 - p50 is auxiliary
             p50 = None
         result = {
-            "metric": "rccl-tests all_reduce bus-bw (GB/s)",
+            "metric": f"rccl-tests {args.op} bus-bw (GB/s)",
             "value": round(busbw, 2),
             "unit": "GB/s",
             "n_gpus": n,
@@ -313,8 +342,8 @@ def main():
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": "all_reduce_perf",
-                "collective": "all_reduce",
+                "model": f"{args.op}_perf",
+                "collective": args.op,
                 "message_bytes": args.max_bytes,
                 "mode": mode,
                 "global_batch": None,
