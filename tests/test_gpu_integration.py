@@ -352,3 +352,20 @@ def test_metrics_http_endpoint_on_hardware():
     finally:
         smi.shutdown()
         amdsmi.ops = None
+
+
+def test_gpu_doctor_on_hardware():
+    """The triage CLI must report a fully healthy node through the real
+    amdsmi shim (with a short event drain)."""
+    import subprocess
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "cmd", "gpu_doctor.py"),
+         "--drain-events-ms", "100"],
+        env=env, capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "node HEALTHY" in r.stdout
+    assert "[FAIL]" not in r.stdout

@@ -286,3 +286,24 @@ def test_extra_mounts_only_when_host_path_exists(tmp_path):
     assert len(mounts) == 2
     assert mounts[1]["container_path"] == "/etc/OpenCL/vendors"
     assert mounts[1]["read_only"] is True
+
+
+def test_gpu_doctor_mock(tmp_path, capsys):
+    """gpu_doctor triage over the mock backend: healthy node exits 0; a
+    driverless node (no kfd) exits 1."""
+    import importlib.util
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        "gpu_doctor", os.path.join(repo, "cmd", "gpu_doctor.py"))
+    doctor = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(doctor)
+
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, 2)
+    assert doctor.main(["--mock-amdsmi", "--dev-directory", dev]) == 0
+    out = capsys.readouterr().out
+    assert "node HEALTHY" in out and out.count("[  ok]") >= 8
+
+    empty = str(tmp_path / "nodev")
+    os.makedirs(empty)
+    assert doctor.main(["--mock-amdsmi", "--dev-directory", empty]) == 1
