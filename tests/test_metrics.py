@@ -113,3 +113,45 @@ def test_collector_unknown_device(tmp_path):
     stats = m.AmdSmiCollector().collect(["amdgpu0", "amdgpu9"])
     assert "amdgpu0" in stats and "amdgpu9" not in stats
     assert stats["amdgpu0"]["model"] == "AMD Instinct MI355X"
+
+
+def test_metrics_http_endpoint_cpu(tmp_path):
+    """Full /metrics HTTP scrape on CPU (mock amdsmi + fake pod-resources):
+    node AND container gauges served with the reference's label sets."""
+    import socket
+    import urllib.request
+
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, 2)
+    mock = MockAmdSmi(dev)
+    mock.init()
+    amdsmi.set_ops(mock)
+    mgr = AmdGPUManager(GPUConfig(), dev_directory=dev,
+                        sysfs_root=str(tmp_path / "sys"))
+    mgr.start()
+
+    sock_path = os.path.join(str(tmp_path), "podres.sock")
+    fake = FakePodResources(sock_path, [
+        ("nsA", "pod1", "train", "amd.com/gpu", ["amdgpu0"]),
+    ])
+    fake.start()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    srv = m.MetricServer(
+        mgr, port=port, collection_interval_s=3600,
+        pod_resources=m.PodResourcesClient(sock_path))
+    srv.start()
+    try:
+        srv.collect_once()
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+        assert 'duty_cycle_gpu_node{' in body
+        assert 'make="amd"' in body
+        assert 'pod="pod1"' in body and 'container="train"' in body
+        assert 'request{' in body
+    finally:
+        srv.stop()
+        fake.stop()
+        mgr.stop()
