@@ -120,3 +120,25 @@ def test_configmap_references_resolve():
             walk(d)
     missing = referenced - defined
     assert not missing, f"configMaps referenced but never defined: {missing}"
+
+
+def test_configmap_shell_scripts_parse():
+    """Every *.sh key embedded in a ConfigMap must pass `bash -n`."""
+    import subprocess
+    import tempfile
+    checked = 0
+    for path, items in load_all().items():
+        for d in items:
+            if d.get("kind") != "ConfigMap":
+                continue
+            for key, val in (d.get("data") or {}).items():
+                if not key.endswith(".sh"):
+                    continue
+                with tempfile.NamedTemporaryFile("w", suffix=".sh") as f:
+                    f.write(val)
+                    f.flush()
+                    r = subprocess.run(["bash", "-n", f.name],
+                                       capture_output=True, text=True)
+                    assert r.returncode == 0, f"{path}:{key}: {r.stderr}"
+                checked += 1
+    assert checked >= 3, f"expected >=3 embedded scripts, found {checked}"
