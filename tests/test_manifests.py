@@ -142,3 +142,27 @@ def test_configmap_shell_scripts_parse():
                     assert r.returncode == 0, f"{path}:{key}: {r.stderr}"
                 checked += 1
     assert checked >= 3, f"expected >=3 embedded scripts, found {checked}"
+
+
+def test_device_plugin_args_match_entrypoint():
+    """Every CLI arg the DaemonSet passes must be accepted by
+    cmd/amd_gpu.py's argparse (catches flag drift)."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "amd_gpu_flags", os.path.join(REPO, "cmd", "amd_gpu.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+
+    path = os.path.join(REPO, "deploy", "device-plugin",
+                        "amd-gpu-device-plugin.yaml")
+    with open(path) as f:
+        docs = [d for d in yaml.safe_load_all(f) if d]
+    args = []
+    for d in docs:
+        if d.get("kind") != "DaemonSet":
+            continue
+        for c in d["spec"]["template"]["spec"].get("containers", []):
+            args += [a for a in c.get("args", []) if a.startswith("--")]
+    assert args, "no args found in the device-plugin DaemonSet"
+    parsed = mod.parse_args(args)  # raises SystemExit on unknown flags
+    assert parsed.enable_health_monitoring
