@@ -162,7 +162,8 @@ def test_device_plugin_args_match_entrypoint():
         if d.get("kind") != "DaemonSet":
             continue
         for c in d["spec"]["template"]["spec"].get("containers", []):
-            args += [a for a in c.get("args", []) if a.startswith("--")]
+            argv = list(c.get("command", [])) + list(c.get("args", []))
+            args += [a for a in argv if a.startswith("--")]
     assert args, "no args found in the device-plugin DaemonSet"
     parsed = mod.parse_args(args)  # raises SystemExit on unknown flags
     assert parsed.enable_health_monitoring
