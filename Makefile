@@ -11,7 +11,7 @@ SMI_SO := cea_amd/amdsmi/libceaamd_smi.so
 GPU_SO := cea_amd/ops/libceaamd_gpu.so
 RCCL_BENCH := cea_amd/bin/all_reduce_perf
 
-.PHONY: all smi gpu rcclbench test gputest lint presubmit clean
+.PHONY: all smi gpu rcclbench examples test gputest lint presubmit clean
 
 all: smi gpu rcclbench
 
@@ -44,10 +44,14 @@ lint:
 	python3 -m compileall -q cea_amd cmd tests bench.py __graft_entry__.py
 	python3 build_tools/boilerplate.py
 
-presubmit: lint all test
+examples:
+	$(MAKE) -C example/cu-fencing
+
+presubmit: lint all examples test
 
 clean:
 	rm -f $(SMI_SO) $(GPU_SO) $(RCCL_BENCH)
+	$(MAKE) -C example/cu-fencing clean
 
 # container image targets — parity /root/reference/Makefile:46-93
 IMAGE ?= cea-amd/gpu-device-plugin
