@@ -28,6 +28,9 @@ log = logging.getLogger(__name__)
 MESSAGE_TYPE_REQUEST = 0x1
 MESSAGE_TYPE_RESPONSE = 0x2
 HEADER = struct.Struct(">IIBB")
+# containerd ttrpc caps messages at 4 MiB (channel.go messageLengthMax);
+# anything larger is a protocol violation, not a big message.
+MAX_FRAME_BYTES = 4 << 20
 
 # --- ttrpc Request/Response message types (runtime-built) ------------------
 _pool = descriptor_pool.DescriptorPool()
@@ -145,6 +148,11 @@ class TtrpcEndpoint:
             if hdr is None:
                 break
             length, stream_id, mtype, _flags = HEADER.unpack(hdr)
+            if length > MAX_FRAME_BYTES:
+                # protocol violation (containerd ttrpc caps messages at
+                # 4 MiB); don't attempt an unbounded read — drop the link
+                log.error("ttrpc frame length %d exceeds cap; closing", length)
+                break
             body = self._recv_exact(length) if length else b""
             if body is None:
                 break
@@ -155,7 +163,12 @@ class TtrpcEndpoint:
             elif mtype == MESSAGE_TYPE_RESPONSE:
                 ev = self._pending.pop(stream_id, None)
                 if ev is not None:
-                    self._results[stream_id] = Response.FromString(body)
+                    try:
+                        self._results[stream_id] = Response.FromString(body)
+                    except Exception as e:  # noqa: BLE001 - malformed peer data
+                        self._results[stream_id] = Response(
+                            status=Status(code=13,
+                                          message=f"malformed response: {e}"))
                     ev.set()
         self._closed.set()
         # wake all pending callers so they fail fast instead of timing out

@@ -173,3 +173,50 @@ def test_ttrpc_wire_format():
     t.join(timeout=2)
     ep.close()
     b.close()
+
+
+def test_ttrpc_rejects_oversized_and_malformed_frames():
+    """Wire robustness: a frame claiming >4 MiB drops the link (no unbounded
+    read); a malformed Response body fails the pending call with a status
+    instead of killing the read loop."""
+    import socket
+    import threading
+    import time
+
+    from cea_amd.nri import ttrpc as t
+
+    # oversized frame: peer sends a huge length; endpoint must close
+    a, b = socket.socketpair()
+    ep = t.TtrpcEndpoint(a)
+    ep.start()
+    b.sendall(t.HEADER.pack(t.MAX_FRAME_BYTES + 1, 1, t.MESSAGE_TYPE_REQUEST, 0))
+    deadline = time.time() + 5
+    while not ep._closed.is_set() and time.time() < deadline:
+        time.sleep(0.02)
+    assert ep._closed.is_set()
+    ep.close()
+    b.close()
+
+    # malformed response: pending call gets an error status, loop survives
+    a, b = socket.socketpair()
+    ep = t.TtrpcEndpoint(a)
+    ep.start()
+    result = {}
+
+    def caller():
+        try:
+            ep.call("svc", "M", b"", timeout=5)
+        except t.TtrpcError as e:
+            result["err"] = e
+
+    th = threading.Thread(target=caller)
+    th.start()
+    time.sleep(0.2)
+    # stream id 1 is the first client call; reply with garbage protobuf
+    b.sendall(t.HEADER.pack(3, 1, t.MESSAGE_TYPE_RESPONSE, 0) + b"\xff\xff\xff")
+    th.join(timeout=5)
+    assert not th.is_alive()
+    assert "err" in result and result["err"].code == 13
+    assert not ep._closed.is_set()  # read loop still alive
+    ep.close()
+    b.close()
