@@ -45,3 +45,14 @@ def test_all_reduce_sweep_with_check():
 def test_other_collectives(op):
     out = run_bench(["-o", op])
     assert "# done" in out
+
+
+def test_graph_mode_and_dtypes():
+    """-G hipGraph replay and -d datatypes (bf16/fp8) must complete the
+    sweep (nccl-tests CLI parity features)."""
+    out = run_bench(["-G", "1", "-c", "1"])
+    assert "# done" in out
+    out = run_bench(["-d", "bf16"])
+    assert "dtype=bf16" in out and "# done" in out
+    out = run_bench(["-d", "fp8e4m3"])
+    assert "# done" in out
