@@ -9,7 +9,7 @@ RUN apt-get update && apt-get install -y --no-install-recommends \
       python3 python3-pip && \
     rm -rf /var/lib/apt/lists/* && \
     pip3 install --no-cache-dir grpcio protobuf prometheus_client \
-      pyyaml requests
+      pyyaml requests fastapi uvicorn
 
 COPY Makefile ./
 COPY csrc/ csrc/
