@@ -48,14 +48,19 @@ __global__ void vector_add_vec4(const float4* __restrict__ a,
   }
 }
 
-// Fault path: same shape as the reference demo kernel — every thread indexes
-// far out of bounds, raising a GPU page fault (KFD VMFAULT event) that the
-// health checker must observe.
+// Fault path: same purpose as the reference demo kernel (vectorAdd.cu:28-34,
+// +1e12 index) — deliberately write out of bounds so the KFD raises a VM
+// fault the health checker must observe.  The offset is +1 GiB, NOT +1e12
+// elements: a multi-TiB offset lands outside the GPU VA aperture on some
+// ROCm builds and is classified HSA_STATUS_ERROR_MEMORY_APERTURE_VIOLATION
+// instead of a page fault, which does NOT emit the amdsmi VMFAULT event.
+// +1 GiB stays inside the aperture but beyond any backing -> true VM page
+// fault (AMDSMI_EVT_NOTIF_VMFAULT / PAGE_FAULT_START) on every build.
 __global__ void vector_add_oob(const float* a, const float* b, float* c,
                                long n) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) {
-    c[i + 1000000000000L] = a[i] + b[i];
+    c[i + (1L << 28)] = a[i] + b[i];  // +2^28 floats = +1 GiB
   }
 }
 
