@@ -345,9 +345,9 @@ class AmdGPUManager:
             service = PluginService(self)
             # Allocate() is the latency-critical RPC (pure in-memory map
             # lookups, parity manager.go Allocate path); bias grpc for
-            # latency.  Measured on MI355X node: p50 ≈ 1.0 ms with a
-            # python-grpc client (the floor of the python round trip; the
-            # handler itself adds <200 µs over a no-op RPC).
+            # latency.  Measured on an MI355X box: p50 ≈ 220 µs end-to-end
+            # with a python-grpc client — the python-grpc round-trip floor;
+            # the handler itself is ≈7 µs (profiles/BENCH_n1_r01c.json).
             server = grpc.server(
                 futures.ThreadPoolExecutor(max_workers=8),
                 options=[("grpc.optimization_target", "latency")],
