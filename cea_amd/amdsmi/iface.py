@@ -3,7 +3,7 @@
 Mirrors the shape of the reference's NvmlOperations seam
 (/root/reference/pkg/gpu/nvidia/nvmlutil/nvmlutil.go:30-37) but with
 AMD-native semantics: render-node minors instead of nvidia minors, BDF
-instead of PciInfo, SPX/DPX/CPX compute-partition info instead of MIG mode,
+instead of PciInfo, SPX/DPX/QPX/CPX compute-partition info instead of MIG mode,
 RAS/ECC/thermal events instead of Xids.
 """
 from __future__ import annotations

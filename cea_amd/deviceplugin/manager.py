@@ -6,7 +6,7 @@ for MI355X:
     nvidiactl+nvidia-uvm defaultDevices, manager.go:378-388) plus one
     /dev/dri/renderD<minor> per allocated GPU (vs /dev/nvidiaN),
   * discovery goes through the AmdSmiOperations seam (vs NVML),
-  * partitioning is SPX/DPX/CPX render-node enumeration (vs MIG capability
+  * partitioning is SPX/DPX/QPX/CPX render-node enumeration (vs MIG capability
     files),
   * the MPS strategy maps to CU-mask env fencing (cu-fencing) since ROCm has
     no MPS daemon.
