@@ -103,9 +103,15 @@ main() {
     verify_installation
     exit 0
   fi
-  install_kernel_headers
-  setup_repos
-  build_and_load_kmd
+  if [[ "${SKIP_KMD_BUILD:-0}" == "1" ]]; then
+    # preloaded-KMD nodes (daemonset-preloaded.yaml): amdgpu is baked into
+    # the image; only stage the userspace.
+    setup_repos
+  else
+    install_kernel_headers
+    setup_repos
+    build_and_load_kmd
+  fi
   install_rocm_userspace
   verify_installation
   update_host_ld_cache
