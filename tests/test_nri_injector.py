@@ -158,8 +158,12 @@ def test_ttrpc_wire_format():
     u8 type, u8 flags (channel.go:63-90)."""
     a, b = socket.socketpair(socket.AF_UNIX, socket.SOCK_STREAM)
     ep = TtrpcEndpoint(a)
+    ep.start()
+    result = {}
     t = threading.Thread(
-        target=lambda: ep.call("svc", "M", b"hello", timeout=2), daemon=True
+        target=lambda: result.update(
+            out=ep.call("svc", "M", b"hello", timeout=5)),
+        daemon=True,
     )
     t.start()
     hdr = b.recv(10)
@@ -170,7 +174,9 @@ def test_ttrpc_wire_format():
     assert mtype == MESSAGE_TYPE_REQUEST and flags == 0 and stream_id % 2 == 1
     resp = Response(payload=b"world").SerializeToString()
     b.sendall(HEADER.pack(len(resp), stream_id, MESSAGE_TYPE_RESPONSE, 0) + resp)
-    t.join(timeout=2)
+    t.join(timeout=5)
+    assert not t.is_alive()
+    assert result.get("out") == b"world"
     ep.close()
     b.close()
 
