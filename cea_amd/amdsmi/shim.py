@@ -62,13 +62,23 @@ def load_library() -> ctypes.CDLL:
 
 
 class ShimAmdSmi(AmdSmiOperations):
-    # amdsmi event-notification mask (AMDSMI_EVENT_MASK_FROM_INDEX(i) =
-    # 1<<(i-1)): VMFAULT(1) | THERMAL_THROTTLE(2) | GPU_PRE_RESET(3) |
-    # GPU_POST_RESET(4) | PAGE_FAULT_START(7) | PAGE_FAULT_END(8) — the
-    # health-relevant subset of amdsmi_evt_notification_type_t.
-    DEFAULT_EVENT_MASK = (
+    # Fallback event-notification mask, used only when the loaded shim .so
+    # predates cea_smi_default_event_mask (AMDSMI_EVENT_MASK_FROM_INDEX(i)
+    # = 1<<(i-1)): VMFAULT(1) | THERMAL_THROTTLE(2) | GPU_PRE_RESET(3) |
+    # GPU_POST_RESET(4) | PAGE_FAULT_START(7) | PAGE_FAULT_END(8).  The
+    # authoritative mask comes from the shim, which builds it from the
+    # amdsmi header it was compiled against — enum indices differ across
+    # amdsmi versions, so Python must not hardcode them.
+    FALLBACK_EVENT_MASK = (
         (1 << 0) | (1 << 1) | (1 << 2) | (1 << 3) | (1 << 6) | (1 << 7)
     )
+
+    def default_event_mask(self) -> int:
+        fn = getattr(self.lib, "cea_smi_default_event_mask", None)
+        if fn is None:
+            return self.FALLBACK_EVENT_MASK
+        fn.restype = ctypes.c_ulonglong
+        return int(fn())
 
     def __init__(self, sampler_interval_ms: int = 160):
         self.lib = load_library()
@@ -197,7 +207,7 @@ class ShimAmdSmi(AmdSmiOperations):
         if not self._events_inited:
             self._check(
                 self.lib.cea_smi_event_init(
-                    ctypes.c_ulonglong(self.DEFAULT_EVENT_MASK)
+                    ctypes.c_ulonglong(self.default_event_mask())
                 ),
                 "event_init",
             )

@@ -91,6 +91,13 @@ class GPUHealthChecker:
         self._ecc_baseline: Dict[int, int] = {}
         self._xgmi_state: Dict[int, int] = {}
         self._condition_lock = threading.Lock()
+        # Checker-local record of devices this checker marked Unhealthy and
+        # the fault codes that did it.  Recovery keys off THIS, not
+        # manager.device_health: the manager map is only populated when a
+        # ListAndWatch stream drains the health queue, so relying on it made
+        # recovery ordering-dependent when kubelet was disconnected
+        # (ADVICE r01 low).
+        self._unhealthy: Dict[str, Set[int]] = {}
 
     # -- lifecycle -----------------------------------------------------------
     def start(self) -> None:
@@ -198,35 +205,103 @@ class GPUHealthChecker:
         affected = self._affected_device_ids(ev)
         for dev_id in affected:
             log.warning("marking device %s Unhealthy (event %d)", dev_id, ev.code)
+            self._unhealthy.setdefault(dev_id, set()).add(ev.code)
             self.manager.health.put(api.Device(ID=dev_id, health=api.UNHEALTHY))
 
     def _recover_after_reset(self, ev: Event) -> None:
-        """GPU_POST_RESET: if the device answers a probe again, re-mark the
-        partitions the reset touched Healthy — kubelet restores capacity
-        without a plugin restart."""
+        """GPU_POST_RESET: re-mark the partitions of the *matched* die
+        Healthy, but only after (a) the device answers a probe again and
+        (b) every persistent fault class that made it Unhealthy re-verifies
+        clean.  Unlike catch_error, an empty or unmatched event UUID
+        recovers NOTHING: the all-devices fallback is right when marking
+        Unhealthy (never drop a node-critical signal over an id-format
+        mismatch) and wrong for recovery (never return a persistently-bad
+        device to the kubelet pool on a vague signal)."""
         ops = amdsmi.get_ops()
-        affected = self._affected_device_ids(ev)
-        for dev_id in affected:
-            if self.manager.device_health.get(dev_id) != api.UNHEALTHY:
+        matched = self._matched_device_ids(ev)
+        if not matched:
+            log.warning(
+                "post-reset event uuid %r matches no enumerated device; "
+                "not recovering anything", ev.device_uuid)
+            return
+        for dev_id in matched:
+            codes = set(self._unhealthy.get(dev_id, ()))
+            if not codes and self.manager.device_health.get(dev_id) != api.UNHEALTHY:
+                continue
+            idx = self._device_index(dev_id)
+            if idx is None:
                 continue
             try:
                 # probe through the same seam discovery uses
-                idx = self._device_index(dev_id)
-                if idx is None:
-                    continue
                 ops.memory_info(idx)
             except Exception as e:  # noqa: BLE001 - still broken, stay Unhealthy
                 log.warning("post-reset probe failed for %s: %s", dev_id, e)
                 continue
+            if not self._fault_classes_clean(dev_id, idx, codes, ops):
+                continue
             log.warning("device %s recovered after GPU reset; marking Healthy",
                         dev_id)
+            self._unhealthy.pop(dev_id, None)
             self.manager.health.put(api.Device(ID=dev_id, health=api.HEALTHY))
+
+    def _fault_classes_clean(self, dev_id: str, idx: int, codes: Set[int],
+                             ops) -> bool:
+        """Re-verify the persistent fault classes (uncorrectable ECC, xGMI
+        link error) before declaring a reset successful.  A reset is
+        expected to clear the fault; if the counter is still above the
+        recorded baseline or the link still reports errors, the fault
+        survived the reset and the device stays Unhealthy."""
+        if EVT_ECC_UNCORRECTABLE in codes:
+            try:
+                count = ops.ecc_uncorrectable_count(idx)
+            except Exception as e:  # noqa: BLE001
+                log.warning("post-reset ECC re-check failed for %s: %s",
+                            dev_id, e)
+                return False
+            if count > self._ecc_baseline.get(idx, 0):
+                log.warning(
+                    "device %s still reports uncorrectable ECC after reset "
+                    "(%d > baseline %d); staying Unhealthy",
+                    dev_id, count, self._ecc_baseline.get(idx, 0))
+                return False
+            # counters may have been cleared by the reset; resync the poll
+            # baseline so the watchdog does not re-fire on the old value
+            self._ecc_baseline[idx] = count
+        if EVT_XGMI_ERROR in codes:
+            try:
+                xgmi = ops.xgmi_error_status(idx)
+            except Exception as e:  # noqa: BLE001
+                log.warning("post-reset xGMI re-check failed for %s: %s",
+                            dev_id, e)
+                return False
+            if xgmi != 0:
+                log.warning("device %s xGMI link still in error state %d "
+                            "after reset; staying Unhealthy", dev_id, xgmi)
+                return False
+            self._xgmi_state[idx] = 0
+        return True
 
     def _device_index(self, dev_id: str):
         pm = self.manager.partition_manager
         infos = pm.devices if pm else self.manager.device_infos
         info = infos.get(dev_id)
         return info.index if info is not None else None
+
+    def _matched_device_ids(self, ev: Event):
+        """Device ids whose die UUID matches the event's, or [] when the
+        UUID is empty or unmatched.  No all-devices fallback here — callers
+        choose their own policy for the vague cases."""
+        if not ev.device_uuid:
+            return []
+        pm = self.manager.partition_manager
+        if pm:
+            return pm.devices_for_die_uuid(ev.device_uuid)
+        uuid = ev.device_uuid.strip().lower()
+        return [
+            dev_id
+            for dev_id, info in self.manager.device_infos.items()
+            if info.uuid.strip().lower() == uuid
+        ]
 
     def _affected_device_ids(self, ev: Event):
         """No UUID => every device (parity health_checker.go:415-424); with a
@@ -240,15 +315,7 @@ class GPUHealthChecker:
                    else list(self.manager.devices.keys()))
         if not ev.device_uuid:
             return all_ids
-        uuid = ev.device_uuid.strip().lower()
-        if pm:
-            matched = pm.devices_for_die_uuid(ev.device_uuid)
-        else:
-            matched = [
-                dev_id
-                for dev_id, info in self.manager.device_infos.items()
-                if info.uuid.strip().lower() == uuid
-            ]
+        matched = self._matched_device_ids(ev)
         if not matched:
             log.warning(
                 "event uuid %s matches no enumerated device; marking all",

@@ -400,6 +400,27 @@ int cea_smi_lib_version(char* buf, int len) {
 
 // --- event notification --------------------------------------------------
 
+// Default notification mask built from THIS header's enum values, so the
+// armed bits always agree with the amdsmi library the shim is compiled
+// against (the driver installer builds the shim on the node, against the
+// same ROCm it installs).  Hardcoding bit indices in Python silently armed
+// the wrong events on amdsmi builds whose amdsmi_evt_notification_type_t
+// numbers differ (e.g. trees where index 5 is RING_HANG instead of
+// MIGRATE_START) — ADVICE r01.  If an enumerator is renamed/removed this
+// fails loudly at compile time, which is the behavior we want.
+unsigned long long cea_smi_default_event_mask(void) {
+  return AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_VMFAULT) |
+         AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_THERMAL_THROTTLE) |
+         AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_GPU_PRE_RESET) |
+         AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_GPU_POST_RESET) |
+         AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_PAGE_FAULT_START) |
+         AMDSMI_EVENT_MASK_FROM_INDEX(AMDSMI_EVT_NOTIF_PAGE_FAULT_END);
+  // MIGRATE_START/END (5/6) and QUEUE_EVICTION/RESTORE (9/10) are benign
+  // traffic and deliberately not armed.  On amdsmi builds that add a
+  // RING_HANG enumerator, extend this expression — the compile-time name
+  // reference keeps the gating tied to the actual header version.
+}
+
 int cea_smi_event_init(unsigned long long mask) {
   std::lock_guard<std::mutex> lk(g_mutex);
   int rc = 0;

@@ -245,3 +245,85 @@ def test_post_reset_recovers_unhealthy_device(tmp_path):
     # a healthy device's reset event pushes nothing
     hc.catch_error(Event(device_uuid="mock-uuid-0", code=EVT_GPU_POST_RESET))
     assert drain(mgr.health) == []
+
+
+def test_post_reset_unmatched_uuid_recovers_nothing(tmp_path):
+    """Recovery is strictly UUID-matched: a GPU_POST_RESET with an empty or
+    unmatched UUID must NOT re-mark Unhealthy devices Healthy (the
+    all-devices fallback is for *marking* Unhealthy only) — otherwise a
+    vague reset signal returns persistently-bad devices to the pool
+    (ADVICE r01 medium)."""
+    from cea_amd.amdsmi.iface import EVT_GPU_POST_RESET
+
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_ECC_UNCORRECTABLE))
+    drain(mgr.health)
+
+    hc.catch_error(Event(device_uuid="", code=EVT_GPU_POST_RESET))
+    assert drain(mgr.health) == []
+    hc.catch_error(Event(device_uuid="GPU-UNKNOWN-FORMAT",
+                         code=EVT_GPU_POST_RESET))
+    assert drain(mgr.health) == []
+    # a correctly-attributed reset still recovers it
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    assert [d.ID for d in drain(mgr.health)] == ["amdgpu1"]
+
+
+def test_post_reset_keeps_device_with_persistent_ecc(tmp_path):
+    """A device whose uncorrectable-ECC counter is still above baseline
+    after the reset stays Unhealthy: the reset did not clear the fault."""
+    from cea_amd.amdsmi.iface import EVT_GPU_POST_RESET
+
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc._ecc_baseline = {0: 0, 1: 0}
+    mock.set_ecc_count(1, 3)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_ECC_UNCORRECTABLE))
+    drain(mgr.health)
+
+    # counter still elevated after the reset => stays Unhealthy
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    assert drain(mgr.health) == []
+
+    # reset actually cleared the counter => recovers, baseline resynced
+    mock.set_ecc_count(1, 0)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    recovered = drain(mgr.health)
+    assert [d.ID for d in recovered] == ["amdgpu1"]
+    assert recovered[0].health == "Healthy"
+    assert hc._ecc_baseline[1] == 0
+
+
+def test_post_reset_keeps_device_with_persistent_xgmi_error(tmp_path):
+    """Same for xGMI: link still in error state after reset => Unhealthy."""
+    from cea_amd.amdsmi.iface import EVT_GPU_POST_RESET, EVT_XGMI_ERROR
+
+    config = GPUConfig(health_critical_events={EVT_XGMI_ERROR})
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2, config=config)
+    mock.set_xgmi_status(1, 2)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_XGMI_ERROR))
+    assert [d.ID for d in drain(mgr.health)] == ["amdgpu1"]
+
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    assert drain(mgr.health) == []
+
+    mock.set_xgmi_status(1, 0)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    assert [d.ID for d in drain(mgr.health)] == ["amdgpu1"]
+    assert hc._xgmi_state.get(1, 0) == 0
+
+
+def test_post_reset_recovery_without_listandwatch_drain(tmp_path):
+    """Recovery must work even when kubelet never drained the health queue
+    (manager.device_health empty): the checker tracks its own Unhealthy
+    set."""
+    from cea_amd.amdsmi.iface import EVT_GPU_POST_RESET
+
+    mgr, mock, kube, hc = make_env(tmp_path, num_gpus=2)
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_ECC_UNCORRECTABLE))
+    # deliberately do NOT drain/sync mgr.device_health (kubelet disconnected);
+    # the Unhealthy entry is still queued
+    assert mgr.device_health.get("amdgpu1") is None
+
+    hc.catch_error(Event(device_uuid="mock-uuid-1", code=EVT_GPU_POST_RESET))
+    msgs = drain(mgr.health)
+    assert msgs[-1].ID == "amdgpu1" and msgs[-1].health == "Healthy"
