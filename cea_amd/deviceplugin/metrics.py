@@ -170,7 +170,10 @@ class MetricServer:
         self._last_reset = time.monotonic()
 
     def start(self) -> None:
-        start_http_server(self.port)
+        # start_http_server returns (WSGIServer, Thread); keep them so
+        # stop() can actually close the listening socket — without this a
+        # restart inside one process leaks the port binding.
+        self._httpd, self._http_thread = start_http_server(self.port)
         self._thread = threading.Thread(target=self._loop, daemon=True)
         self._thread.start()
         log.info("metrics server on :%d (collect every %.0fs)",
@@ -178,6 +181,18 @@ class MetricServer:
 
     def stop(self) -> None:
         self._stop.set()
+        httpd = getattr(self, "_httpd", None)
+        if httpd is not None:
+            try:
+                httpd.shutdown()
+                httpd.server_close()
+            except Exception as e:  # noqa: BLE001
+                log.error("metrics HTTP listener close failed: %s", e)
+            self._httpd = None
+        t = getattr(self, "_http_thread", None)
+        if t is not None:
+            t.join(timeout=5)
+            self._http_thread = None
 
     def _loop(self) -> None:
         while not self._stop.wait(self.collection_interval_s):

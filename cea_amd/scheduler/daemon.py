@@ -23,12 +23,22 @@ import time
 from collections import defaultdict
 from typing import Dict, List, Optional
 
-from .topology import CandidateNode, assign_pods, topo_key_from_labels
+from .topology import (
+    CandidateNode,
+    HeteroNode,
+    assign_pods,
+    assign_pods_hetero,
+    selector_matches,
+    topo_key_from_labels,
+)
 
 log = logging.getLogger(__name__)
 
 GATE_PREFIX = "gke.io/topology-aware-auto-"   # parity schedule-daemon.py:760
 COMPLETION_INDEX_LABEL = "batch.kubernetes.io/job-completion-index"
+# kubeflow training-operator pods carry their gang index here instead
+# (parity schedule-daemon.py:59,373-377)
+KUBEFLOW_REPLICA_INDEX_LABEL = "training.kubeflow.org/replica-index"
 JOB_NAME_LABELS = (
     "job-name",
     "jobset.sigs.k8s.io/jobset-name",
@@ -88,17 +98,23 @@ def job_key(pod: dict) -> str:
     return f"{ns}/{meta.get('name', '')}"
 
 
-def completion_index(pod: dict) -> int:
-    """Parity :119-150 — job completion index label, else trailing ordinal
-    in the pod name."""
+def completion_index(pod: dict):
+    """Sorting key, parity :119-150 — job completion-index label, then the
+    kubeflow replica-index label, else (name-prefix, trailing ordinal) so
+    "xxx-pod2" sorts before "xxx-pod10" (pod_sorting_key semantics).
+    Always returns a (prefix, int) tuple so keys compare consistently."""
     labels = pod.get("metadata", {}).get("labels", {}) or {}
-    if COMPLETION_INDEX_LABEL in labels:
-        try:
-            return int(labels[COMPLETION_INDEX_LABEL])
-        except ValueError:
-            pass
-    m = re.search(r"(\d+)$", pod.get("metadata", {}).get("name", ""))
-    return int(m.group(1)) if m else 0
+    for lab in (COMPLETION_INDEX_LABEL, KUBEFLOW_REPLICA_INDEX_LABEL):
+        if lab in labels:
+            try:
+                return ("", int(labels[lab]))
+            except ValueError:
+                pass
+    name = pod.get("metadata", {}).get("name", "")
+    m = re.fullmatch(r"(.*?)(\d+)", name)
+    if m:
+        return (m.group(1), int(m.group(2)))
+    return (name, 0)
 
 
 def node_is_ready(node: dict) -> bool:
@@ -219,24 +235,8 @@ class TopologyScheduler:
                 continue
             self._last_attempt[key] = now
             job_pods.sort(key=completion_index)
-            req = pod_requests(job_pods[0])
-            candidates = []
-            for n in nodes:
-                if not tolerates(job_pods[0], n):
-                    continue
-                free = node_free_resources(n, pods_by_node[n["metadata"]["name"]])
-                cap = pods_fit_count(free, req)
-                if cap > 0:
-                    candidates.append(CandidateNode(
-                        name=n["metadata"]["name"],
-                        topo=topo_key_from_labels(
-                            n["metadata"].get("labels", {}) or {}),
-                        capacity=cap,
-                    ))
-            assignment = assign_pods(len(job_pods), candidates)
+            assignment = self._assign_job(job_pods, nodes, pods_by_node, key)
             if assignment is None:
-                log.info("job %s: %d pods do not fit on %d candidate nodes; "
-                         "waiting", key, len(job_pods), len(candidates))
                 continue
             for pod, node_name in zip(job_pods, assignment):
                 try:
@@ -248,12 +248,102 @@ class TopologyScheduler:
                               pod["metadata"].get("name"), e)
         return bound
 
-    def run_forever(self) -> None:
-        log.info("topology scheduler loop starting (interval %.0fs)",
-                 self.interval_s)
+    def _assign_job(self, job_pods: List[dict], nodes: List[dict],
+                    pods_by_node: Dict[str, List[dict]], key: str
+                    ) -> Optional[List[str]]:
+        """Per-pod resource extraction (parity :356-417) with the fast
+        provably-optimal window path for homogeneous gangs and the greedy
+        hetero path otherwise.  Per-pod nodeSelector (parity :421-436) and
+        taint toleration are checked in both paths."""
+        reqs = [pod_requests(p) for p in job_pods]
+        selectors = [
+            p.get("spec", {}).get("nodeSelector") or {} for p in job_pods
+        ]
+        homogeneous = (
+            all(r == reqs[0] for r in reqs[1:])
+            and all(s == selectors[0] for s in selectors[1:])
+            and all(
+                (p.get("spec", {}).get("tolerations") or [])
+                == (job_pods[0].get("spec", {}).get("tolerations") or [])
+                for p in job_pods[1:]
+            )
+        )
+        free_by_node = {
+            n["metadata"]["name"]: node_free_resources(
+                n, pods_by_node[n["metadata"]["name"]])
+            for n in nodes
+        }
+        if homogeneous:
+            candidates = []
+            for n in nodes:
+                labels = n["metadata"].get("labels", {}) or {}
+                if not tolerates(job_pods[0], n):
+                    continue
+                if not selector_matches(selectors[0], labels):
+                    continue
+                cap = pods_fit_count(free_by_node[n["metadata"]["name"]],
+                                     reqs[0])
+                if cap > 0:
+                    candidates.append(CandidateNode(
+                        name=n["metadata"]["name"],
+                        topo=topo_key_from_labels(labels),
+                        capacity=cap,
+                    ))
+            assignment = assign_pods(len(job_pods), candidates)
+            if assignment is None:
+                log.info("job %s: %d pods do not fit on %d candidate nodes; "
+                         "waiting", key, len(job_pods), len(candidates))
+            return assignment
+        # heterogeneous gang: per-pod requests/selectors differ
+        hetero_pods = []
+        for pod, req, sel in zip(job_pods, reqs, selectors):
+            hetero_pods.append({"requests": req, "node_selector": sel,
+                                "_pod": pod})
+        hetero_nodes = []
+        for n in nodes:
+            # toleration must hold for every pod that could land here; check
+            # per-pod by folding it into the selector path is not possible,
+            # so conservatively require all gang pods tolerate the node
+            if not all(tolerates(p, n) for p in job_pods):
+                continue
+            hetero_nodes.append(HeteroNode(
+                name=n["metadata"]["name"],
+                topo=topo_key_from_labels(
+                    n["metadata"].get("labels", {}) or {}),
+                free=free_by_node[n["metadata"]["name"]],
+                labels=n["metadata"].get("labels", {}) or {},
+            ))
+        assignment = assign_pods_hetero(hetero_pods, hetero_nodes)
+        if assignment is None:
+            log.info("job %s (heterogeneous): %d pods do not fit on %d "
+                     "candidate nodes; waiting", key, len(job_pods),
+                     len(hetero_nodes))
+        return assignment
+
+    def run_forever(self, warmup_s: float = 90.0, settle_s: float = 5.0
+                    ) -> None:
+        """Parity with the reference loop cadence (:777-807): a 90 s warmup
+        before the first pass (so pods scheduled by a previous daemon
+        incarnation become visible on nodes and count against free
+        resources), then a settle delay once gated pods are seen (so all
+        members of a gang are visible before placement)."""
+        log.info("topology scheduler loop starting (interval %.0fs, "
+                 "warmup %.0fs)", self.interval_s, warmup_s)
+        if warmup_s > 0:
+            time.sleep(warmup_s)
         while not self._stop:
             try:
+                if settle_s > 0 and self._has_gated_pending():
+                    # gang members may still be appearing; let them settle
+                    time.sleep(settle_s)
                 self.schedule_once()
             except Exception as e:  # noqa: BLE001
                 log.error("scheduling pass failed: %s", e)
             time.sleep(self.interval_s)
+
+    def _has_gated_pending(self) -> bool:
+        try:
+            pods = self.kube.list_pods(field_selector="status.phase=Pending")
+        except Exception:  # noqa: BLE001
+            return False
+        return any(has_topology_gate(p) for p in pods)

@@ -82,31 +82,122 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
     topology distance.  Returns one node name per pod (nodes repeat up to
     their capacity), or None if the gang does not fit.
 
-    Nodes are sorted by topology key; every contiguous window with enough
-    capacity is scored by the summed pairwise distance of the slots it
-    provides; best window wins.
-    """
+    Exact for homogeneous gangs: expand nodes into a sorted slot list
+    (node repeated per capacity), then slide a `num_pods`-wide window over
+    EVERY slot offset with O(1) incremental score updates.  An optimal
+    slot set is always contiguous under the hierarchical metric: for
+    sorted slots i<=j<=k, dist(i,k) >= dist(j,k) and >= dist(i,j), so any
+    selected extreme slot can be moved inward toward the rest without
+    increasing any pairwise term (compaction argument).  Enumerating all
+    slot offsets (not just node-aligned starts) also covers windows that
+    take partial capacity of their first node.  O(S) overall with
+    S = sum(min(capacity, num_pods)) slots, vs the reference's exhaustive
+    itertools-style search (schedule-daemon.py:500-544)."""
     nodes = [n for n in nodes if n.capacity > 0]
     nodes.sort(key=lambda n: (n.topo.sort_key(), n.name))
-    total_cap = sum(n.capacity for n in nodes)
-    if total_cap < num_pods or num_pods <= 0:
+    if num_pods <= 0:
+        return None
+    slots: List[CandidateNode] = []
+    for n in nodes:
+        slots.extend([n] * min(n.capacity, num_pods))
+    if len(slots) < num_pods:
         return None
 
+    # sliding window state: per-level value counts and same-pair sums
+    counts = [dict(), dict(), dict()]
+    same = [0, 0, 0]     # sum of c*(c-1)/2 per level
+    levels = (lambda t: t.block, lambda t: t.subblock, lambda t: t.host)
+
+    def add(t: TopoKey, sign: int) -> None:
+        for li, keyfn in enumerate(levels):
+            k = keyfn(t)
+            c = counts[li].get(k, 0)
+            same[li] -= c * (c - 1) // 2
+            c += sign
+            counts[li][k] = c
+            same[li] += c * (c - 1) // 2
+
+    g = num_pods
+    total_pairs = g * (g - 1) // 2
+    weights = (W_BLOCK, W_SUBBLOCK, W_HOST)
+    best_score = None
+    best_start = 0
+    for i in range(g):
+        add(slots[i].topo, +1)
+    start = 0
+    while True:
+        score = sum(w * (total_pairs - s) for w, s in zip(weights, same))
+        if best_score is None or score < best_score:
+            best_score, best_start = score, start
+            if score == 0:
+                break
+        if start + g >= len(slots):
+            break
+        add(slots[start].topo, -1)
+        add(slots[start + g].topo, +1)
+        start += 1
+    return [s.name for s in slots[best_start:best_start + g]]
+
+
+@dataclasses.dataclass
+class HeteroNode:
+    name: str
+    topo: TopoKey
+    free: Dict[str, float]
+    labels: Dict[str, str] = dataclasses.field(default_factory=dict)
+
+
+def selector_matches(selector: Optional[Dict[str, str]],
+                     labels: Dict[str, str]) -> bool:
+    """pod.spec.nodeSelector semantics: every key must match exactly
+    (parity: can_schedule, schedule-daemon.py:421-436)."""
+    return all(labels.get(k) == v for k, v in (selector or {}).items())
+
+
+def assign_pods_hetero(pods: List[dict], nodes: List[HeteroNode]
+                       ) -> Optional[List[str]]:
+    """Heterogeneous gangs (per-pod requests / nodeSelector differ, the
+    case the reference handles via exponential search at
+    schedule-daemon.py:500-544): contiguous-window greedy.  For each
+    start position in topology-sorted order, first-fit each pod (in gang
+    order) onto the first node >= start with enough remaining resources
+    and a matching nodeSelector; the packed placement is scored by summed
+    pairwise distance and the best window wins.  Correct and
+    topology-aware; optimality is only guaranteed on the homogeneous path
+    (assign_pods).
+
+    Each pod dict: {"requests": {res: qty}, "node_selector": {k: v}}.
+    Returns one node name per pod in input order, or None."""
+    nodes = sorted(nodes, key=lambda n: (n.topo.sort_key(), n.name))
     best: Optional[Tuple[int, List[str]]] = None
     for start in range(len(nodes)):
-        slots: List[str] = []
+        remaining = [dict(n.free) for n in nodes]
+        placement: List[str] = []
         topos: List[TopoKey] = []
-        for n in nodes[start:]:
-            take = min(n.capacity, num_pods - len(slots))
-            slots.extend([n.name] * take)
-            topos.extend([n.topo] * take)
-            if len(slots) == num_pods:
+        ok = True
+        for pod in pods:
+            placed = False
+            for ni in range(start, len(nodes)):
+                node, rem = nodes[ni], remaining[ni]
+                if not selector_matches(pod.get("node_selector"), node.labels):
+                    continue
+                reqs = pod.get("requests", {})
+                if any(rem.get(k, 0.0) < v for k, v in reqs.items() if v > 0):
+                    continue
+                for k, v in reqs.items():
+                    rem[k] = rem.get(k, 0.0) - v
+                placement.append(node.name)
+                topos.append(node.topo)
+                placed = True
                 break
-        if len(slots) < num_pods:
-            break  # windows further right only get smaller
+            if not placed:
+                ok = False
+                break
+        if not ok:
+            continue
         score = _pairwise_score(topos)
         if best is None or score < best[0]:
-            best = (score, slots)
+            best = (score, placement)
         if best[0] == 0:
             break
     return best[1] if best else None

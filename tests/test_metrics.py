@@ -155,3 +155,43 @@ def test_metrics_http_endpoint_cpu(tmp_path):
         srv.stop()
         fake.stop()
         mgr.stop()
+
+
+def test_metric_server_stop_releases_port(tmp_path):
+    """stop() must close the HTTP listener so the same port can be rebound
+    in-process (prometheus start_http_server is
+    fire-and-forget; a restart leaked the binding)."""
+    import socket
+    import urllib.request
+    import urllib.error
+
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, 1)
+    mock = MockAmdSmi(dev)
+    mock.init()
+    amdsmi.set_ops(mock)
+    mgr = AmdGPUManager(GPUConfig(), dev_directory=dev,
+                        sysfs_root=str(tmp_path / "sys"))
+    mgr.start()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    srv = m.MetricServer(mgr, port=port, collection_interval_s=3600)
+    srv.start()
+    urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics", timeout=10)
+    srv.stop()
+    with pytest.raises((urllib.error.URLError, ConnectionError, OSError)):
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics", timeout=2)
+
+    # restart on the SAME port succeeds because the listener was closed
+    srv2 = m.MetricServer(mgr, port=port, collection_interval_s=3600)
+    srv2.start()
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+        assert "duty_cycle" in body or body is not None
+    finally:
+        srv2.stop()
+        mgr.stop()

@@ -113,3 +113,88 @@ def test_preferred_allocation_invariants(avail_keys, n_must, size):
     # must_include honored up to size
     for m in must[:size]:
         assert m in got
+
+
+@given(
+    st.integers(1, 5),
+    st.lists(
+        st.tuples(st.integers(0, 2), st.integers(0, 1), st.integers(0, 2),
+                  st.integers(1, 3)),
+        min_size=1,
+        max_size=6,
+    ),
+)
+@settings(max_examples=60, deadline=None)
+def test_assign_pods_matches_exhaustive_optimum(num_pods, node_spec):
+    """The contiguous-window placement achieves the same minimal summed
+    pairwise distance as exhaustive search over ALL slot combinations —
+    the optimality claim in topology.py verified end-to-end on assign_pods
+    itself (previously only the scoring function was
+    brute-force-checked)."""
+    import itertools
+
+    nodes = [
+        topo.CandidateNode(f"n{i}", topo.TopoKey(f"b{b}", f"s{s}", f"h{h}"),
+                           cap)
+        for i, (b, s, h, cap) in enumerate(node_spec)
+    ]
+    slots = []
+    for n in nodes:
+        slots.extend([n.topo] * n.capacity)
+    got = topo.assign_pods(num_pods, nodes)
+    if len(slots) < num_pods:
+        assert got is None
+        return
+    assert got is not None and len(got) == num_pods
+    topo_by_name = {n.name: n.topo for n in nodes}
+    got_score = topo._pairwise_score([topo_by_name[name] for name in got])
+    brute = min(
+        topo._pairwise_score([slots[i] for i in combo])
+        for combo in itertools.combinations(range(len(slots)), num_pods)
+    )
+    assert got_score == brute, (got_score, brute, node_spec, num_pods)
+
+
+@given(
+    st.lists(
+        st.tuples(st.integers(1, 4), st.booleans()),  # (gpu req, wants label)
+        min_size=1, max_size=5,
+    ),
+    st.lists(
+        st.tuples(st.integers(0, 2), st.integers(2, 8), st.booleans()),
+        min_size=1, max_size=5,
+    ),
+)
+@settings(max_examples=60, deadline=None)
+def test_assign_pods_hetero_respects_resources_and_selectors(pod_spec,
+                                                             node_spec):
+    """Heterogeneous placement never oversubscribes a node and never
+    violates a pod's nodeSelector."""
+    from collections import defaultdict
+
+    pods = [
+        {"requests": {"amd.com/gpu": float(g)},
+         "node_selector": ({"pool": "special"} if sel else {})}
+        for g, sel in pod_spec
+    ]
+    nodes = [
+        topo.HeteroNode(
+            name=f"n{i}",
+            topo=topo.TopoKey(f"b{b}", "s", f"h{i}"),
+            free={"amd.com/gpu": float(cap)},
+            labels=({"pool": "special"} if lab else {}),
+        )
+        for i, (b, cap, lab) in enumerate(node_spec)
+    ]
+    got = topo.assign_pods_hetero(pods, nodes)
+    if got is None:
+        return
+    assert len(got) == len(pods)
+    by_name = {n.name: n for n in nodes}
+    used = defaultdict(float)
+    for pod, name in zip(pods, got):
+        node = by_name[name]
+        assert topo.selector_matches(pod["node_selector"], node.labels)
+        used[name] += pod["requests"]["amd.com/gpu"]
+    for name, total in used.items():
+        assert total <= by_name[name].free["amd.com/gpu"]

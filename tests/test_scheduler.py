@@ -176,8 +176,16 @@ def test_taints_respected():
 
 
 def test_completion_index_order():
-    assert d.completion_index(make_pod("x", idx=7)) == 7
-    assert d.completion_index(make_pod("worker-12")) == 12
+    assert d.completion_index(make_pod("x", idx=7)) == ("", 7)
+    assert d.completion_index(make_pod("worker-12")) == ("worker-", 12)
+    # kubeflow replica-index label honored (parity schedule-daemon.py:373-377)
+    kf = make_pod("kf")
+    kf["metadata"]["labels"][d.KUBEFLOW_REPLICA_INDEX_LABEL] = "3"
+    assert d.completion_index(kf) == ("", 3)
+    # natural numeric ordering: pod2 before pod10
+    pods = [make_pod("w-pod10"), make_pod("w-pod2")]
+    pods.sort(key=d.completion_index)
+    assert [p["metadata"]["name"] for p in pods] == ["w-pod2", "w-pod10"]
 
 
 def test_job_grouping():
@@ -239,3 +247,95 @@ def test_pairwise_score_matches_bruteforce():
                     for i in range(len(topos))
                     for j in range(i + 1, len(topos)))
         assert topo._pairwise_score(topos) == brute
+
+
+def test_node_selector_feasibility():
+    """A pod's spec.nodeSelector restricts candidate nodes (parity
+    schedule-daemon.py:421-436) — even when a non-matching node is
+    topologically closer."""
+    kube = FakeKubeClient(nodes=[
+        make_node("plain1", topo_labels("b1", "s1", "h1")),
+        make_node("plain2", topo_labels("b1", "s1", "h1")),
+        make_node("labeled", {**topo_labels("b9", "s9", "h9"),
+                              "pool": "mi355x"}),
+    ])
+    p = make_pod("sel-0", idx=0)
+    p["spec"]["nodeSelector"] = {"pool": "mi355x"}
+    kube.pods = {("default", "sel-0"): p}
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 1
+    bound = kube.get_pod("default", "sel-0")
+    vals = bound["spec"]["affinity"]["nodeAffinity"][
+        "requiredDuringSchedulingIgnoredDuringExecution"
+    ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"]
+    assert vals == ["labeled"]
+
+
+def test_heterogeneous_gang_per_pod_resources():
+    """Pods of one gang with different GPU requests are placed with their
+    OWN requests (job_pods[0] was used for every pod).  A
+    6-GPU and a 2-GPU pod pack into one 8-GPU node — something the
+    homogeneous model could never see."""
+    kube = FakeKubeClient(nodes=[
+        make_node("packed", topo_labels("b1", "s1", "h1"), gpus=8),
+        make_node("spare", topo_labels("b2", "s2", "h2"), gpus=8),
+    ])
+    big = make_pod("het-0", idx=0, gpus=6)
+    small = make_pod("het-1", idx=1, gpus=2)
+    kube.pods = {("default", "het-0"): big, ("default", "het-1"): small}
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 2
+    for name in ("het-0", "het-1"):
+        bound = kube.get_pod("default", name)
+        vals = bound["spec"]["affinity"]["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"
+        ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"]
+        assert vals == ["packed"], (name, vals)
+
+
+def test_heterogeneous_gang_does_not_oversubscribe():
+    """Two 6-GPU pods cannot share an 8-GPU node; gang spans two nodes."""
+    kube = FakeKubeClient(nodes=[
+        make_node("n1", topo_labels("b1", "s1", "h1"), gpus=8),
+        make_node("n2", topo_labels("b1", "s1", "h2"), gpus=8),
+    ])
+    a = make_pod("two-0", idx=0, gpus=6)
+    b = make_pod("two-1", idx=1, gpus=6)
+    # make requests differ slightly so the hetero path is taken
+    b["spec"]["containers"][0]["resources"]["requests"]["cpu"] = "3"
+    kube.pods = {("default", "two-0"): a, ("default", "two-1"): b}
+    sched = d.TopologyScheduler(kube, gate_cooloff_s=0)
+    assert sched.schedule_once() == 2
+    bound_nodes = set()
+    for name in ("two-0", "two-1"):
+        bound = kube.get_pod("default", name)
+        vals = bound["spec"]["affinity"]["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"
+        ]["nodeSelectorTerms"][0]["matchExpressions"][0]["values"]
+        bound_nodes.add(vals[0])
+    assert bound_nodes == {"n1", "n2"}
+
+
+def test_run_forever_warmup_delays_first_pass():
+    """run_forever waits warmup_s before the first pass (parity
+    schedule-daemon.py:777-807: racing a deployment's pod creation can
+    split a gang)."""
+    import threading
+    import time as _t
+
+    kube = FakeKubeClient(nodes=[make_node("n1", topo_labels("b", "s", "h"))])
+    kube.pods = {("default", "w-0"): make_pod("w-0", idx=0)}
+    sched = d.TopologyScheduler(kube, interval_s=0.05, gate_cooloff_s=0)
+    t = threading.Thread(
+        target=sched.run_forever, kwargs={"warmup_s": 0.5, "settle_s": 0},
+        daemon=True)
+    t.start()
+    _t.sleep(0.2)
+    # still in warmup: nothing bound yet
+    assert d.has_topology_gate(kube.get_pod("default", "w-0"))
+    deadline = _t.time() + 5
+    while _t.time() < deadline and d.has_topology_gate(
+            kube.get_pod("default", "w-0")):
+        _t.sleep(0.05)
+    sched._stop = True
+    assert not d.has_topology_gate(kube.get_pod("default", "w-0"))
