@@ -31,6 +31,7 @@ import json
 import os
 import statistics
 import time
+import sys
 
 
 def parse_args():
@@ -234,12 +235,23 @@ def main():
                 for _ in range(3):
                     step(buf_send, buf_recv)
             torch.cuda.current_stream().wait_stream(side)
-            torch.cuda.synchronize()
             # Let the ProcessGroupNCCL watchdog dequeue the completed
             # warmup works before capture starts: its hipEventQuery on a
             # still-enqueued work during capture is
-            # hipErrorStreamCaptureUnsupported and aborts the process.
-            time.sleep(0.25)
+            # hipErrorStreamCaptureUnsupported and aborts the process
+            # (a watchdog-thread abort cannot be caught by the except
+            # below).  The watchdog sweeps its work list every ~100 ms,
+            # so poll in slices — synchronize each slice so every work's
+            # event is complete when the watchdog looks — instead of one
+            # fixed sleep; bound is tunable for loaded nodes, and
+            # CEA_BENCH_GRAPH=0 disables capture entirely.
+            drain_ms = int(os.environ.get("CEA_BENCH_GRAPH_DRAIN_MS", "600"))
+            deadline = time.monotonic() + drain_ms / 1000.0
+            while True:
+                torch.cuda.synchronize()
+                if time.monotonic() >= deadline:
+                    break
+                time.sleep(0.1)
             g = torch.cuda.CUDAGraph()
             # thread_local: event queries from OTHER threads (the NCCL
             # watchdog) stay legal while this thread captures.
@@ -247,7 +259,11 @@ def main():
                 for _ in range(iters_per_graph):
                     step(buf_send, buf_recv)
             return g.replay, True
-        except Exception:  # noqa: BLE001 - capture support is optional
+        except Exception as e:  # noqa: BLE001 - capture support is optional
+            print(f"hipGraph capture failed ({e}); falling back to eager. "
+                  "Set CEA_BENCH_GRAPH=0 to skip capture, or raise "
+                  "CEA_BENCH_GRAPH_DRAIN_MS on loaded nodes.",
+                  file=sys.stderr)
             torch.cuda.synchronize()
 
             def eager():
