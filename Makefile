@@ -10,12 +10,21 @@ CXX ?= g++
 SMI_SO := cea_amd/amdsmi/libceaamd_smi.so
 GPU_SO := cea_amd/ops/libceaamd_gpu.so
 RCCL_BENCH := cea_amd/bin/all_reduce_perf
+FAKE_SMI_SO := tests/_build/libamd_smi.so
 
-.PHONY: all smi gpu rcclbench examples test gputest lint presubmit clean
+.PHONY: all smi gpu rcclbench fake-smi examples test gputest lint presubmit clean
 
 all: smi gpu rcclbench
 
 smi: $(SMI_SO)
+
+# test-only fake libamd_smi, LD_PRELOADed so CPU tests drive the REAL shim
+fake-smi: $(FAKE_SMI_SO)
+
+$(FAKE_SMI_SO): csrc/fake_amdsmi.cpp
+	mkdir -p tests/_build
+	$(CXX) -O2 -std=c++17 -Wall -shared -fPIC -pthread \
+	  -I$(ROCM)/include $< -o $@
 
 $(SMI_SO): csrc/amdsmi_shim.cpp
 	$(CXX) -O2 -std=c++17 -Wall -shared -fPIC -pthread \
