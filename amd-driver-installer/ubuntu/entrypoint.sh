@@ -17,6 +17,9 @@ set -x
 
 ROCM_VERSION="${ROCM_VERSION:-7.2}"
 AMDGPU_DRIVER_VERSION="${AMDGPU_DRIVER_VERSION:-30.20}"
+# install prefix of the ROCm userspace to stage from (versioned prefixes
+# like /opt/rocm-7.2.0 exist on some images)
+ROCM_DIR="${ROCM_DIR:-/opt/rocm}"
 ROOT_MOUNT_DIR="${ROOT_MOUNT_DIR:-/root}"
 INSTALL_DIR_HOST="${INSTALL_DIR_HOST:-/home/kubernetes/bin/amd}"
 INSTALL_DIR="${ROOT_MOUNT_DIR}${INSTALL_DIR_HOST}"
@@ -71,17 +74,22 @@ build_and_load_kmd() {
 }
 
 install_rocm_userspace() {
-  DEBIAN_FRONTEND=noninteractive apt-get install -y \
-    rocm-core rocminfo rocm-smi-lib amd-smi-lib \
-    hip-runtime-amd rocblas hipblaslt miopen-hip rccl
+  if [[ "${SKIP_PACKAGE_INSTALL:-0}" != "1" ]]; then
+    DEBIAN_FRONTEND=noninteractive apt-get install -y \
+      rocm-core rocminfo rocm-smi-lib amd-smi-lib \
+      hip-runtime-amd rocblas hipblaslt miopen-hip rccl
+  fi
+  # SKIP_PACKAGE_INSTALL=1: the installer image already ships the ROCm
+  # userspace at /opt/rocm (preloaded installer images, air-gapped
+  # clusters); stage it directly without touching apt.
   mkdir -p "${INSTALL_DIR}/lib64" "${INSTALL_DIR}/bin" "${INSTALL_DIR}/.info"
   # The subset GPU pods need, mounted read-only by the device plugin
   # (manager.mounts(): /home/kubernetes/bin/amd -> /usr/local/amd).
-  cp -a /opt/rocm/lib/*.so* "${INSTALL_DIR}/lib64/" 2>/dev/null || true
+  cp -a "${ROCM_DIR}"/lib/*.so* "${INSTALL_DIR}/lib64/" 2>/dev/null || true
   for tool in rocminfo rocm-smi amd-smi hipconfig; do
-    [[ -x "/opt/rocm/bin/${tool}" ]] && cp -a "/opt/rocm/bin/${tool}" "${INSTALL_DIR}/bin/"
+    [[ -x "${ROCM_DIR}/bin/${tool}" ]] && cp -a "${ROCM_DIR}/bin/${tool}" "${INSTALL_DIR}/bin/"
   done
-  cp -a /opt/rocm/.info/version "${INSTALL_DIR}/.info/version" 2>/dev/null || true
+  cp -a "${ROCM_DIR}/.info/version" "${INSTALL_DIR}/.info/version" 2>/dev/null || true
 }
 
 verify_installation() {
@@ -101,12 +109,12 @@ update_host_ld_cache() {
 main() {
   if check_cached_version; then
     verify_installation
-    exit 0
+    return 0
   fi
   if [[ "${SKIP_KMD_BUILD:-0}" == "1" ]]; then
     # preloaded-KMD nodes (daemonset-preloaded.yaml): amdgpu is baked into
     # the image; only stage the userspace.
-    setup_repos
+    [[ "${SKIP_PACKAGE_INSTALL:-0}" == "1" ]] || setup_repos
   else
     install_kernel_headers
     setup_repos
@@ -119,4 +127,8 @@ main() {
   echo "amdgpu + ROCm install complete"
 }
 
-main "$@"
+# Source-guard: tests source this file and drive the functions directly
+# against a scratch ROOT_MOUNT_DIR (tests/test_installer_exec.py).
+if [[ "${BASH_SOURCE[0]:-}" == "$0" ]]; then
+  main "$@"
+fi

@@ -1,0 +1,153 @@
+"""EXECUTES the ubuntu driver-installer entrypoint (not just `bash -n`).
+
+The reference's installers are fleet-proven; round 1 only parse-checked
+ours.  Here the real script runs in this ROCm container
+with ROOT_MOUNT_DIR pointed at a scratch root, SKIP_KMD_BUILD=1 (no
+kernel to build against) and SKIP_PACKAGE_INSTALL=1 (no network; the
+image's /opt/rocm is the userspace source, the same path the preloaded
+installer image uses in production), asserting the staged tree layout
+the device plugin mounts (manager: /home/kubernetes/bin/amd ->
+/usr/local/amd).  Parity model:
+/root/reference/nvidia-driver-installer/ubuntu/entrypoint.sh:33-163.
+"""
+import os
+import subprocess
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+ENTRYPOINT = os.path.join(REPO, "amd-driver-installer", "ubuntu",
+                          "entrypoint.sh")
+
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir("/opt/rocm/lib"),
+    reason="needs a ROCm userspace at /opt/rocm to stage from",
+)
+
+
+def run_installer(root, extra="main", env=None, rocm_dir=None):
+    """Source the entrypoint and run `extra` (a bash snippet) against a
+    scratch root."""
+    e = dict(os.environ)
+    e.update({
+        "ROOT_MOUNT_DIR": str(root),
+        "SKIP_KMD_BUILD": "1",
+        "SKIP_PACKAGE_INSTALL": "1",
+    })
+    if rocm_dir:
+        e["ROCM_DIR"] = str(rocm_dir)
+    e.update(env or {})
+    script = f". '{ENTRYPOINT}'\n{extra}\n"
+    return subprocess.run(["bash", "-c", script], env=e,
+                          capture_output=True, text=True, timeout=300)
+
+
+@pytest.fixture()
+def mini_rocm(tmp_path):
+    """A small stand-in for /opt/rocm so the staging copy is instant (the
+    real tree is multi-GB); the real-inventory check below asserts the
+    actual /opt/rocm ships what staging would pick up."""
+    r = tmp_path / "rocm"
+    os.makedirs(r / "lib")
+    os.makedirs(r / "bin")
+    os.makedirs(r / ".info")
+    for lib in ("librccl.so.1.0", "libamd_smi.so.25", "libamdhip64.so.7",
+                "librocm_smi64.so.7"):
+        (r / "lib" / lib).write_bytes(b"\x7fELF-stub")
+        base = lib.split(".so")[0] + ".so"
+        os.symlink(lib, r / "lib" / base)
+    for tool in ("rocminfo", "rocm-smi", "amd-smi", "hipconfig"):
+        p = r / "bin" / tool
+        p.write_text("#!/bin/sh\necho stub\n")
+        p.chmod(0o755)
+    (r / ".info" / "version").write_text("7.2.0-stub\n")
+    return r
+
+
+def make_fake_host(root):
+    """Device nodes + etc the verify step expects on a real node."""
+    os.makedirs(root / "dev" / "dri", exist_ok=True)
+    (root / "dev" / "kfd").write_bytes(b"")
+    (root / "dev" / "dri" / "renderD128").write_bytes(b"")
+    os.makedirs(root / "etc" / "ld.so.conf.d", exist_ok=True)
+
+
+def test_stage_userspace_layout(tmp_path, mini_rocm):
+    """install_rocm_userspace stages lib64 + bin + .info from ROCM_DIR
+    into ROOT_MOUNT_DIR/home/kubernetes/bin/amd, preserving symlinks."""
+    proc = run_installer(tmp_path, "install_rocm_userspace",
+                         rocm_dir=mini_rocm)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    install = tmp_path / "home" / "kubernetes" / "bin" / "amd"
+    libs = os.listdir(install / "lib64")
+    assert any(l.startswith("librccl.so") for l in libs), sorted(libs)[:20]
+    assert any(l.startswith("libamd_smi.so") for l in libs)
+    assert any(l.startswith("libamdhip64.so") for l in libs)
+    # soname symlinks survive cp -a
+    assert os.path.islink(install / "lib64" / "librccl.so")
+    bins = os.listdir(install / "bin")
+    assert "rocminfo" in bins and "amd-smi" in bins
+    assert (install / ".info" / "version").read_text().startswith("7.2")
+
+
+def test_real_rocm_ships_required_inventory():
+    """The actual /opt/rocm in this image contains everything staging
+    picks up — the libraries GPU pods depend on (RCCL transport, SMI for
+    health, HIP runtime) and the verify tools."""
+    libs = os.listdir("/opt/rocm/lib")
+    for want in ("librccl.so", "libamd_smi.so", "libamdhip64.so"):
+        assert any(l.startswith(want) for l in libs), want
+    for tool in ("rocminfo", "amd-smi"):
+        assert os.path.exists(f"/opt/rocm/bin/{tool}"), tool
+
+
+def test_full_main_flow_and_cache_idempotency(tmp_path, mini_rocm):
+    """main() end-to-end on a fake host root: staged tree, host
+    ld.so.conf, cache file; a second run exits early via the cache
+    (parity: the reference's cache-file idempotency, entrypoint.sh:33-51)."""
+    make_fake_host(tmp_path)
+    # device verification needs real hardware; stub only that function
+    stub = ("verify_installation() { echo VERIFY_STUBBED; }\n"
+            "main")
+    proc = run_installer(tmp_path, stub, rocm_dir=mini_rocm)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "amdgpu + ROCm install complete" in proc.stdout
+
+    install = tmp_path / "home" / "kubernetes" / "bin" / "amd"
+    cache = (install / ".cache").read_text()
+    kernel = os.uname().release
+    assert f"CACHE_KERNEL_VERSION={kernel}" in cache
+    assert "CACHE_ROCM_VERSION=" in cache
+    ld_conf = (tmp_path / "etc" / "ld.so.conf.d" / "amd.conf").read_text()
+    assert ld_conf.strip() == "/home/kubernetes/bin/amd/lib64"
+    # ldconfig -r built a cache inside the scratch root
+    assert (tmp_path / "etc" / "ld.so.cache").exists()
+
+    # second run: cache hit short-circuits before staging
+    proc2 = run_installer(tmp_path, stub, rocm_dir=mini_rocm)
+    assert proc2.returncode == 0, proc2.stderr[-2000:]
+    assert "already installed" in proc2.stdout
+    assert "install complete" not in proc2.stdout
+
+
+def test_cache_invalidated_by_version_change(tmp_path, mini_rocm):
+    """Changing the pinned driver version busts the cache and re-stages."""
+    make_fake_host(tmp_path)
+    stub = "verify_installation() { :; }\nmain"
+    assert run_installer(tmp_path, stub, rocm_dir=mini_rocm).returncode == 0
+    proc = run_installer(tmp_path, stub, rocm_dir=mini_rocm,
+                         env={"AMDGPU_DRIVER_VERSION": "31.0"})
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "install complete" in proc.stdout
+    cache = (tmp_path / "home" / "kubernetes" / "bin" / "amd" /
+             ".cache").read_text()
+    assert "CACHE_AMDGPU_VERSION=31.0" in cache
+
+
+def test_verify_fails_without_device_nodes(tmp_path):
+    """verify_installation is strict: no /dev/kfd under the root => fail
+    (the retry loop in the DaemonSet depends on this being loud)."""
+    os.makedirs(tmp_path / "dev", exist_ok=True)
+    proc = run_installer(tmp_path, "verify_installation")
+    assert proc.returncode != 0
+    assert "kfd missing" in proc.stdout + proc.stderr
