@@ -63,11 +63,31 @@ clean:
 	$(MAKE) -C example/cu-fencing clean
 
 # container image targets — parity /root/reference/Makefile:46-93
+# (the reference builds amd64+arm64 presubmit images; MI355X nodes are
+# amd64-only, but CI cross-build parity is kept via buildx)
 IMAGE ?= cea-amd/gpu-device-plugin
 TAG ?= $(shell cat VERSION 2>/dev/null || echo dev)
+ALL_ARCH ?= amd64 arm64
 
 image:
 	docker build -t $(IMAGE):$(TAG) .
 
 push: image
 	docker push $(IMAGE):$(TAG)
+
+# cross-arch single-platform build (reference: container-% at Makefile:52-60)
+image-%:
+	docker buildx build --platform linux/$* -t $(IMAGE)-$*:$(TAG) --load .
+
+# multi-arch manifest build+push (reference: container-multi-arch + push-multi-arch,
+# Makefile:62-93).  The runtime image is only exercised on amd64 (MI355X);
+# arm64 exists for CI build parity.
+image-multi-arch:
+	docker buildx build \
+	  --platform $(shell echo "$(ALL_ARCH)" | sed 's/ /,linux\//; s/^/linux\//') \
+	  -t $(IMAGE):$(TAG) .
+
+push-multi-arch:
+	docker buildx build --push \
+	  --platform $(shell echo "$(ALL_ARCH)" | sed 's/ /,linux\//; s/^/linux\//') \
+	  -t $(IMAGE):$(TAG) .
