@@ -92,6 +92,48 @@ def parse_partition_status(text: str) -> List[Dict[str, str]]:
     return [d for d in out if "accelerator_partition" in d]
 
 
+def parse_partition_profiles(text: str) -> Dict[str, int]:
+    """Parse `amd-smi partition` ACCELERATOR_PARTITION_PROFILES output into
+    {accelerator_type: num_partitions} — the hardware's own statement of
+    which modes it supports and how many logical devices each produces.
+    Format captured from a real MI355X (profiles/pool_probe_r02.log):
+
+        ACCELERATOR_PARTITION_PROFILES:
+        GPU_ID  PROFILE_INDEX  MEMORY_PARTITION_CAPS  ACCELERATOR_TYPE  \
+PARTITION_ID  NUM_PARTITIONS  ...
+        0       0              NPS1                   SPX*              \
+0             1               ...
+                1              NPS1                   DPX               \
+N/A           2               ...
+
+    The currently-active profile is marked with ``*``."""
+    out: Dict[str, int] = {}
+    in_profiles = False
+    for line in text.splitlines():
+        s = line.strip()
+        if s.startswith("ACCELERATOR_PARTITION_PROFILES"):
+            in_profiles = True
+            continue
+        if in_profiles and s.endswith(":") and s == s.upper():
+            break  # next section header
+        if not in_profiles or not s:
+            continue
+        m = re.search(
+            r"\b(SPX|DPX|QPX|TPX|CPX)\*?\s+(?:\d+|N/A)\s+(\d+)\b", s)
+        if m:
+            out[m.group(1)] = int(m.group(2))
+    return out
+
+
+def hardware_partition_capabilities(runner: Runner) -> Dict[str, int]:
+    """Query `amd-smi partition` for the supported mode->count table;
+    empty dict when the CLI subcommand is unavailable (older amd-smi)."""
+    rc, out = runner(["amd-smi", "partition"])
+    if rc != 0:
+        return {}
+    return parse_partition_profiles(out)
+
+
 def check_desired(states: List[Dict[str, str]], compute: str, memory: str) -> bool:
     """Uniformity + desired-mode check (parity: checkDesired,
     partition_gpu.go:446-458)."""
@@ -190,6 +232,22 @@ def run(config_path: str = DEFAULT_CONFIG_PATH,
         log.info("GPUs already partitioned as %s/%s; nothing to do",
                  compute, memory)
         return False
+
+    # pre-flight: if the hardware publishes its profile table, confirm the
+    # requested mode is supported and our count table agrees with the
+    # hardware's NUM_PARTITIONS (catches e.g. TPX-capable parts or future
+    # asymmetric profiles before any state is touched)
+    caps = hardware_partition_capabilities(runner)
+    if caps:
+        if compute not in caps:
+            raise PartitionError(
+                f"hardware does not support compute partition {compute}; "
+                f"supported: {sorted(caps)}")
+        if caps[compute] != PARTITION_COUNT.get(compute):
+            raise PartitionError(
+                f"hardware reports {caps[compute]} partitions for {compute}, "
+                f"static table says {PARTITION_COUNT.get(compute)} — "
+                "refusing to proceed with a mismatched device-count model")
 
     # memory partition first (it implies a KFD re-enumeration), then compute
     needs_mem = any(
