@@ -7,11 +7,12 @@ topology keys from node labels (:46-48, :175-195), pairwise distance
 search (:500-544).
 
 Redesign: the reference enumerates ALL sorted node combinations
-(itertools.combinations at :500-544) which is exponential; here nodes are
-sorted by (block, subblock, host) and assignment picks the best contiguous
-window by total pairwise distance — same optimum for hierarchical distances
-(an optimal set under a tree metric is always contiguous in DFS order of the
-tree), O(N·G) instead of O(C(N,G)).
+(itertools.combinations at :500-544) which is exponential; here the
+hierarchical all-pairs objective decomposes per topology-tree element as
+W*C(k,2), so a bottom-up knapsack DP over the block->subblock->host tree
+finds the EXACT optimum in O(G * total_capacity) per level — verified
+equivalent to exhaustive search by hypothesis
+(tests/test_properties.py::test_assign_pods_matches_exhaustive_optimum).
 """
 from __future__ import annotations
 
@@ -57,15 +58,24 @@ def topo_key_from_labels(labels: Dict[str, str]) -> TopoKey:
 
 
 def distance(a: TopoKey, b: TopoKey) -> int:
-    """Pairwise topology distance (parity: the 1e6/1e4/1e2 weighting)."""
-    d = 0
+    """Pairwise topology distance (parity: the 1e6/1e4/1e2 weighting).
+
+    Levels are compared as PATHS (a host belongs to its subblock, a
+    subblock to its block): nodes in different blocks are maximally far
+    even if a lower-level label value coincides.  This equals the
+    reference's bare per-level comparison (schedule-daemon.py:153-172)
+    whenever label values are unique across the fleet — the real-world
+    case — and, unlike bare comparison, it is a true tree metric, which
+    is what makes contiguous-window placement provably optimal
+    (hypothesis found the counterexample for bare labels: a reused host
+    label under a different subblock breaks sorted-order monotonicity)."""
     if a.block != b.block:
-        d += W_BLOCK
+        return W_BLOCK + W_SUBBLOCK + W_HOST
     if a.subblock != b.subblock:
-        d += W_SUBBLOCK
+        return W_SUBBLOCK + W_HOST
     if a.host != b.host:
-        d += W_HOST
-    return d
+        return W_HOST
+    return 0
 
 
 @dataclasses.dataclass
@@ -82,61 +92,125 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
     topology distance.  Returns one node name per pod (nodes repeat up to
     their capacity), or None if the gang does not fit.
 
-    Exact for homogeneous gangs: expand nodes into a sorted slot list
-    (node repeated per capacity), then slide a `num_pods`-wide window over
-    EVERY slot offset with O(1) incremental score updates.  An optimal
-    slot set is always contiguous under the hierarchical metric: for
-    sorted slots i<=j<=k, dist(i,k) >= dist(j,k) and >= dist(i,j), so any
-    selected extreme slot can be moved inward toward the rest without
-    increasing any pairwise term (compaction argument).  Enumerating all
-    slot offsets (not just node-aligned starts) also covers windows that
-    take partial capacity of their first node.  O(S) overall with
-    S = sum(min(capacity, num_pods)) slots, vs the reference's exhaustive
-    itertools-style search (schedule-daemon.py:500-544)."""
-    nodes = [n for n in nodes if n.capacity > 0]
-    nodes.sort(key=lambda n: (n.topo.sort_key(), n.name))
+    EXACT optimum in polynomial time (vs the reference's exponential
+    itertools-style search, schedule-daemon.py:500-544): under the
+    hierarchical path metric, minimizing the summed pairwise distance of
+    G slots equals maximizing the bonus
+
+        sum_hosts W_HOST*C(k_h,2) + sum_subblocks W_SUBBLOCK*C(k_s,2)
+                                  + sum_blocks W_BLOCK*C(k_b,2)
+
+    where k_x = slots taken inside topology element x (score =
+    (W_B+W_S+W_H)*C(G,2) - bonus).  The bonus decomposes over the
+    block->subblock->host tree, so a bottom-up knapsack DP over the tree
+    finds the exact max: each tree node combines its children's value
+    vectors (per-child cost bounded by its subtree capacity, total
+    O(G * sum(capacity)) per level) and adds its own W*C(k,2).
+    Verified equivalent to exhaustive slot-combination search by
+    tests/test_properties.py::test_assign_pods_matches_exhaustive_optimum.
+
+    (An earlier contiguous-window design was NOT optimal here: with
+    all-pairs scoring, skipping an anomalous middle host can beat every
+    window — hypothesis found the counterexample.)"""
     if num_pods <= 0:
         return None
-    slots: List[CandidateNode] = []
-    for n in nodes:
-        slots.extend([n] * min(n.capacity, num_pods))
-    if len(slots) < num_pods:
+    nodes = [n for n in nodes if n.capacity > 0]
+    G = num_pods
+    if sum(n.capacity for n in nodes) < G:
         return None
 
-    # sliding window state: per-level value counts and same-pair sums
-    counts = [dict(), dict(), dict()]
-    same = [0, 0, 0]     # sum of c*(c-1)/2 per level
-    levels = (lambda t: t.block, lambda t: t.subblock, lambda t: t.host)
+    # group nodes by full topology path; build block -> subblock -> host
+    buckets: Dict[Tuple[str, str, str], List[CandidateNode]] = {}
+    for n in nodes:
+        buckets.setdefault((n.topo.block, n.topo.subblock, n.topo.host),
+                           []).append(n)
+    tree: Dict[str, Dict[str, Dict[Tuple[str, str, str], int]]] = {}
+    for path, members in buckets.items():
+        b, s, _h = path
+        cap = min(sum(m.capacity for m in members), G)
+        tree.setdefault(b, {}).setdefault(s, {})[path] = cap
 
-    def add(t: TopoKey, sign: int) -> None:
-        for li, keyfn in enumerate(levels):
-            k = keyfn(t)
-            c = counts[li].get(k, 0)
-            same[li] -= c * (c - 1) // 2
-            c += sign
-            counts[li][k] = c
-            same[li] += c * (c - 1) // 2
+    NEG = float("-inf")
 
-    g = num_pods
-    total_pairs = g * (g - 1) // 2
-    weights = (W_BLOCK, W_SUBBLOCK, W_HOST)
-    best_score = None
-    best_start = 0
-    for i in range(g):
-        add(slots[i].topo, +1)
-    start = 0
-    while True:
-        score = sum(w * (total_pairs - s) for w, s in zip(weights, same))
-        if best_score is None or score < best_score:
-            best_score, best_start = score, start
-            if score == 0:
+    def pairs(k: int) -> int:
+        return k * (k - 1) // 2
+
+    def combine(children, bonus_w):
+        """Knapsack-merge child (vals, rec) pairs; add bonus_w*C(k,2).
+        Returns (vals over 0..G, rec(k) -> {hostpath: count})."""
+        acc = [0] + [NEG] * G
+        acc_cap = 0
+        backs = []
+        for vals, _rec in children:
+            child_cap = len(vals) - 1
+            new_cap = min(acc_cap + child_cap, G)
+            new = [NEG] * (G + 1)
+            back = [0] * (G + 1)
+            for k in range(new_cap + 1):
+                best = NEG
+                bj = 0
+                jlo = max(0, k - acc_cap)
+                for j in range(jlo, min(k, child_cap) + 1):
+                    a = acc[k - j]
+                    if a == NEG:
+                        continue
+                    v = a + vals[j]
+                    if v > best:
+                        best, bj = v, j
+                new[k] = best
+                back[k] = bj
+            backs.append(back)
+            acc = new
+            acc_cap = new_cap
+
+        out_vals = [
+            acc[k] + bonus_w * pairs(k) if acc[k] != NEG else NEG
+            for k in range(G + 1)
+        ]
+
+        def rec(k: int) -> Dict[Tuple[str, str, str], int]:
+            out: Dict[Tuple[str, str, str], int] = {}
+            kk = k
+            for i in range(len(children) - 1, -1, -1):
+                j = backs[i][kk]
+                if j:
+                    out.update(children[i][1](j))
+                kk -= j
+            return out
+
+        return out_vals, rec
+
+    def leaf(path, cap):
+        vals = [W_HOST * pairs(k) for k in range(cap + 1)]
+
+        def rec(k: int) -> Dict[Tuple[str, str, str], int]:
+            return {path: k} if k else {}
+
+        return vals, rec
+
+    block_entries = []
+    for b in sorted(tree):
+        sub_entries = []
+        for s in sorted(tree[b]):
+            hosts = [leaf(path, cap)
+                     for path, cap in sorted(tree[b][s].items())]
+            sub_entries.append(combine(hosts, W_SUBBLOCK))
+        block_entries.append(combine(sub_entries, W_BLOCK))
+    root_vals, root_rec = combine(block_entries, 0)
+    if root_vals[G] == NEG:
+        return None
+    chosen = root_rec(G)
+
+    out: List[str] = []
+    for path in sorted(chosen):
+        k = chosen[path]
+        for n in sorted(buckets[path], key=lambda n: n.name):
+            take = min(n.capacity, k)
+            out.extend([n.name] * take)
+            k -= take
+            if k == 0:
                 break
-        if start + g >= len(slots):
-            break
-        add(slots[start].topo, -1)
-        add(slots[start + g].topo, +1)
-        start += 1
-    return [s.name for s in slots[best_start:best_start + g]]
+    return out
 
 
 @dataclasses.dataclass
@@ -217,8 +291,10 @@ def _pairwise_score(topos: List[TopoKey]) -> int:
             counts[k] = counts.get(k, 0) + 1
         return sum(c * (c - 1) // 2 for c in counts.values())
 
-    # Labels compared exactly as distance() does (bare label equality).
+    # Path keys per level, exactly as distance() compares them.
     score = W_BLOCK * (total_pairs - same_pairs(lambda t: t.block))
-    score += W_SUBBLOCK * (total_pairs - same_pairs(lambda t: t.subblock))
-    score += W_HOST * (total_pairs - same_pairs(lambda t: t.host))
+    score += W_SUBBLOCK * (
+        total_pairs - same_pairs(lambda t: (t.block, t.subblock)))
+    score += W_HOST * (
+        total_pairs - same_pairs(lambda t: (t.block, t.subblock, t.host)))
     return score
