@@ -117,8 +117,11 @@ def main():
     backend = args.backend or ("nccl" if has_gpu else "gloo")
     device = torch.device("cpu")
     if has_gpu:
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # modulo: lets an oversubscription experiment run world>deviceCount
+        # (e.g. 2 ranks on a 1-GPU box); normal runs have world==deviceCount
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29571")
