@@ -133,4 +133,7 @@ main() {
   echo "amdgpu + ROCm install complete for minikube node"
 }
 
-main "$@"
+# Source-guard: tests may source this file and drive functions directly.
+if [[ "${BASH_SOURCE[0]:-}" == "$0" ]]; then
+  main "$@"
+fi

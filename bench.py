@@ -376,6 +376,10 @@ def main():
                 "algbw_GBps": round(algbw, 2),
                 "sweep": sweep,
                 "allocate_p50_us": p50,
+                # RCCL tuning in effect (empty = library defaults; the
+                # shipped recipe is deploy/rccl/rccl-config.yaml)
+                "nccl_env": {k: v for k, v in sorted(os.environ.items())
+                             if k.startswith(("NCCL_", "RCCL_"))},
             },
         }
         print(json.dumps(result))
