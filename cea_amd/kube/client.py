@@ -83,6 +83,39 @@ class KubeClient:
             raise KubeError(r.status_code, r.text[:500])
         return r.json() if r.content else None
 
+    # -- list/watch (informer substrate) -------------------------------------
+    def list_raw(self, path: str, params: Optional[dict] = None) -> dict:
+        """GET a collection, returning the WHOLE list object (metadata.
+        resourceVersion included) — the watch bookmark an informer needs."""
+        return self._req("GET", path, params=params)
+
+    def watch(self, path: str, resource_version: str,
+              params: Optional[dict] = None, timeout_s: int = 300):
+        """Streamed watch: yields (type, object) tuples until the server
+        closes the connection (callers loop + relist on 410 Gone).  Uses
+        the chunked watch protocol: one JSON event per line."""
+        import json as _json
+
+        p = dict(params or {})
+        p.update({
+            "watch": "true",
+            "resourceVersion": resource_version,
+            "allowWatchBookmarks": "true",
+            "timeoutSeconds": str(timeout_s),
+        })
+        r = self.session.get(self.base_url + path, params=p, stream=True,
+                             timeout=timeout_s + 30)
+        if r.status_code >= 400:
+            raise KubeError(r.status_code, r.text[:500])
+        try:
+            for line in r.iter_lines():
+                if not line:
+                    continue
+                ev = _json.loads(line)
+                yield ev.get("type", ""), ev.get("object", {})
+        finally:
+            r.close()
+
     # -- nodes ---------------------------------------------------------------
     def get_node(self, name: str) -> dict:
         return self._req("GET", f"/api/v1/nodes/{name}")

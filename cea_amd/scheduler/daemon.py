@@ -202,16 +202,37 @@ def bind_pod(kube, pod: dict, node_name: str) -> None:
 
 class TopologyScheduler:
     def __init__(self, kube, interval_s: float = 5.0,
-                 gate_cooloff_s: float = 60.0):
+                 gate_cooloff_s: float = 60.0,
+                 pod_informer=None, node_informer=None):
+        """pod_informer/node_informer: optional started Informer caches
+        (cea_amd/kube/informer.py).  When set, passes read memory instead
+        of full-listing the cluster each loop — beyond the reference,
+        which lists per pass (schedule-daemon.py:783)."""
         self.kube = kube
         self.interval_s = interval_s
         self.gate_cooloff_s = gate_cooloff_s   # parity :777-807
+        self.pod_informer = pod_informer
+        self.node_informer = node_informer
         self._last_attempt: Dict[str, float] = {}
         self._stop = False
 
+    def _list_pods_and_pending(self):
+        if self.pod_informer is not None:
+            all_pods = self.pod_informer.items()
+            pending = [p for p in all_pods
+                       if p.get("status", {}).get("phase") == "Pending"]
+            return all_pods, pending
+        pending = self.kube.list_pods(field_selector="status.phase=Pending")
+        return self.kube.list_pods(), pending
+
+    def _list_nodes(self):
+        if self.node_informer is not None:
+            return self.node_informer.items()
+        return self.kube.list_nodes()
+
     def schedule_once(self) -> int:
         """One pass; returns the number of pods bound."""
-        pods = self.kube.list_pods(field_selector="status.phase=Pending")
+        all_pods, pods = self._list_pods_and_pending()
         gated = [p for p in pods if has_topology_gate(p)]
         if not gated:
             return 0
@@ -219,8 +240,7 @@ class TopologyScheduler:
         for p in gated:
             jobs[job_key(p)].append(p)
 
-        nodes = [n for n in self.kube.list_nodes() if node_is_ready(n)]
-        all_pods = self.kube.list_pods()
+        nodes = [n for n in self._list_nodes() if node_is_ready(n)]
         pods_by_node: Dict[str, List[dict]] = defaultdict(list)
         for p in all_pods:
             node_name = p.get("spec", {}).get("nodeName")
