@@ -52,9 +52,42 @@ def rccl_version() -> str:
     return ""
 
 
-def publish_driver_version_annotations(kube_client, node_name: str) -> None:
+def partition_capabilities(sysfs_root: str = "/sys") -> Dict[str, str]:
+    """Partition-mode capability annotations from the KMD's sysfs
+    interface: which compute-partition modes this node's GPUs support
+    (`available_compute_partition`, real format "SPX, DPX, QPX, CPX" —
+    profiles/pool_probe_r02.log) and the current mode.  Operators check
+    these BEFORE applying a gpu_config.json asking for CPX.  Beyond the
+    reference (which has no MIG-capability annotation analog)."""
+    import glob
+
+    out: Dict[str, str] = {}
+    avail: set = set()
+    current: set = set()
+    for path in glob.glob(
+            f"{sysfs_root}/class/drm/card*/device/available_compute_partition"):
+        try:
+            with open(path) as f:
+                avail.update(m.strip().upper()
+                             for m in f.read().strip().split(","))
+            with open(path.replace("available_", "current_")) as f:
+                current.add(f.read().strip().upper())
+        except OSError:
+            continue
+    if avail:
+        out["amd.com/gpu.partition-modes"] = ",".join(sorted(avail - {""}))
+    if len(current) == 1:
+        out["amd.com/gpu.compute-partition"] = next(iter(current))
+    elif current:
+        out["amd.com/gpu.compute-partition"] = "mixed"
+    return out
+
+
+def publish_driver_version_annotations(kube_client, node_name: str,
+                                       sysfs_root: str = "/sys") -> None:
     """Parity: PublishDriverVersionAnnotations (version_visibility.go:38-46);
-    extended with the RCCL release (the transport the L4 layer enables)."""
+    extended with the RCCL release (the transport the L4 layer enables)
+    and the node's partition-mode capabilities."""
     ops = amdsmi.get_ops()
     version = ops.driver_version()
     annotations = parse_version(version)
@@ -62,5 +95,6 @@ def publish_driver_version_annotations(kube_client, node_name: str) -> None:
     rccl = rccl_version()
     if rccl:
         annotations["amd.com/rccl.version"] = rccl
+    annotations.update(partition_capabilities(sysfs_root))
     kube_client.apply_node_annotations(node_name, annotations, FIELD_MANAGER)
     log.info("published driver version annotations: %s", annotations)

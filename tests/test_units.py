@@ -307,3 +307,46 @@ def test_gpu_doctor_mock(tmp_path, capsys):
     empty = str(tmp_path / "nodev")
     os.makedirs(empty)
     assert doctor.main(["--mock-amdsmi", "--dev-directory", empty]) == 1
+
+
+def test_partition_capability_annotations(tmp_path):
+    """Node annotations advertise the KMD's supported partition modes +
+    current mode, using the real sysfs format (pool_probe_r02.log)."""
+    from cea_amd.deviceplugin.version_visibility import (
+        partition_capabilities,
+        publish_driver_version_annotations,
+    )
+
+    sys_root = tmp_path / "sys"
+    d = sys_root / "class" / "drm" / "card0" / "device"
+    os.makedirs(d)
+    (d / "available_compute_partition").write_text("SPX, DPX, QPX, CPX\n")
+    (d / "current_compute_partition").write_text("SPX\n")
+
+    caps = partition_capabilities(str(sys_root))
+    assert caps == {
+        "amd.com/gpu.partition-modes": "CPX,DPX,QPX,SPX",
+        "amd.com/gpu.compute-partition": "SPX",
+    }
+
+    # mixed current modes across cards are flagged
+    d2 = sys_root / "class" / "drm" / "card1" / "device"
+    os.makedirs(d2)
+    (d2 / "available_compute_partition").write_text("SPX, DPX, QPX, CPX\n")
+    (d2 / "current_compute_partition").write_text("CPX\n")
+    caps = partition_capabilities(str(sys_root))
+    assert caps["amd.com/gpu.compute-partition"] == "mixed"
+
+    # end-to-end through the publisher
+    dev = str(tmp_path / "dev")
+    make_fake_dev(dev, 1)
+    mock = MockAmdSmi(dev)
+    mock.init()
+    amdsmi.set_ops(mock)
+    kube = FakeKubeClient(nodes=[{"metadata": {"name": "n1"}, "status": {}}])
+    publish_driver_version_annotations(kube, "n1", sysfs_root=str(sys_root))
+    ann = kube.nodes["n1"]["metadata"]["annotations"]
+    assert ann["amd.com/gpu.partition-modes"] == "CPX,DPX,QPX,SPX"
+
+    # no sysfs knobs (e.g. CPU CI container) => annotations simply absent
+    assert partition_capabilities(str(tmp_path / "nosys")) == {}
