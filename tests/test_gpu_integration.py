@@ -369,3 +369,28 @@ def test_gpu_doctor_on_hardware():
     assert r.returncode == 0, r.stdout + r.stderr
     assert "node HEALTHY" in r.stdout
     assert "[FAIL]" not in r.stdout
+
+
+def test_partition_capability_annotations_on_hardware():
+    """The real sysfs advertises the partition modes; the annotation
+    helper reads them with the real format (r02 feature)."""
+    from cea_amd.deviceplugin.version_visibility import partition_capabilities
+
+    caps = partition_capabilities()
+    assert "amd.com/gpu.partition-modes" in caps, caps
+    modes = caps["amd.com/gpu.partition-modes"].split(",")
+    assert "SPX" in modes and "CPX" in modes, caps
+    assert caps["amd.com/gpu.compute-partition"] in (
+        "SPX", "DPX", "QPX", "CPX"), caps
+
+
+def test_hardware_partition_profiles_real_amd_smi():
+    """`amd-smi partition` on real hardware parses into exactly the
+    mode->count table the whole stack sizes itself from (the r02
+    partitioner pre-flight path, live)."""
+    from cea_amd.partition import partition_gpu as pg
+
+    caps = pg.hardware_partition_capabilities(pg.default_runner)
+    if not caps:
+        pytest.skip("amd-smi partition subcommand unavailable")
+    assert caps == {"SPX": 1, "DPX": 2, "QPX": 4, "CPX": 8}, caps
