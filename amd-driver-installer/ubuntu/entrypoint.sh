@@ -86,9 +86,23 @@ install_rocm_userspace() {
   # The subset GPU pods need, mounted read-only by the device plugin
   # (manager.mounts(): /home/kubernetes/bin/amd -> /usr/local/amd).
   cp -a "${ROCM_DIR}"/lib/*.so* "${INSTALL_DIR}/lib64/" 2>/dev/null || true
+  # ROCm binaries carry RUNPATH $ORIGIN/../lib — a lib->lib64 symlink
+  # makes the staged tree self-contained for them
+  ln -sfn lib64 "${INSTALL_DIR}/lib"
   for tool in rocminfo rocm-smi amd-smi hipconfig; do
     [[ -x "${ROCM_DIR}/bin/${tool}" ]] && cp -a "${ROCM_DIR}/bin/${tool}" "${INSTALL_DIR}/bin/"
   done
+  # amd-smi is a symlink to ../libexec/amdsmi_cli/amdsmi_cli.py (a Python
+  # CLI): stage libexec too or the staged bin/amd-smi dangles
+  if [[ -d "${ROCM_DIR}/libexec/amdsmi_cli" ]]; then
+    mkdir -p "${INSTALL_DIR}/libexec"
+    cp -a "${ROCM_DIR}/libexec/amdsmi_cli" "${INSTALL_DIR}/libexec/"
+  fi
+  # ... and the amdsmi python package the CLI imports
+  if [[ -d "${ROCM_DIR}/share/amd_smi/amdsmi" ]]; then
+    mkdir -p "${INSTALL_DIR}/share/amd_smi"
+    cp -a "${ROCM_DIR}/share/amd_smi/amdsmi" "${INSTALL_DIR}/share/amd_smi/"
+  fi
   cp -a "${ROCM_DIR}/.info/version" "${INSTALL_DIR}/.info/version" 2>/dev/null || true
 }
 

@@ -56,10 +56,19 @@ def mini_rocm(tmp_path):
         (r / "lib" / lib).write_bytes(b"\x7fELF-stub")
         base = lib.split(".so")[0] + ".so"
         os.symlink(lib, r / "lib" / base)
-    for tool in ("rocminfo", "rocm-smi", "amd-smi", "hipconfig"):
+    for tool in ("rocminfo", "rocm-smi", "hipconfig"):
         p = r / "bin" / tool
         p.write_text("#!/bin/sh\necho stub\n")
         p.chmod(0o755)
+    # amd-smi mirrors the real layout: a symlink into libexec (the round-2
+    # staging bug: cp -a of the symlink alone dangles)
+    os.makedirs(r / "libexec" / "amdsmi_cli")
+    cli = r / "libexec" / "amdsmi_cli" / "amdsmi_cli.py"
+    cli.write_text("#!/bin/sh\necho cli-stub\n")
+    cli.chmod(0o755)
+    os.symlink("../libexec/amdsmi_cli/amdsmi_cli.py", r / "bin" / "amd-smi")
+    os.makedirs(r / "share" / "amd_smi" / "amdsmi")
+    (r / "share" / "amd_smi" / "amdsmi" / "__init__.py").write_text("")
     (r / ".info" / "version").write_text("7.2.0-stub\n")
     return r
 
@@ -87,6 +96,13 @@ def test_stage_userspace_layout(tmp_path, mini_rocm):
     assert os.path.islink(install / "lib64" / "librccl.so")
     bins = os.listdir(install / "bin")
     assert "rocminfo" in bins and "amd-smi" in bins
+    # the amd-smi symlink must RESOLVE in the staged tree (libexec staged)
+    assert os.path.exists(install / "bin" / "amd-smi"), "amd-smi dangles"
+    assert (install / "libexec" / "amdsmi_cli" / "amdsmi_cli.py").exists()
+    assert (install / "share" / "amd_smi" / "amdsmi" / "__init__.py").exists()
+    # lib -> lib64 symlink for ROCm RUNPATH ($ORIGIN/../lib) resolution
+    assert os.path.islink(install / "lib")
+    assert (install / "lib" / "librccl.so").exists()
     assert (install / ".info" / "version").read_text().startswith("7.2")
 
 
