@@ -202,6 +202,8 @@ class FakeKubeClient:
         self.events: List[dict] = []
         self.reactors: List[Callable] = []   # (verb, resource, obj) -> maybe raise
         self.lock = threading.Lock()
+        self._rv = 1
+        self._watch_queue: List[tuple] = []
 
     def _react(self, verb: str, resource: str, obj=None):
         for r in self.reactors:
@@ -297,6 +299,47 @@ class FakeKubeClient:
             self._react("create", "events", event)
             self.events.append(copy.deepcopy(event))
             return event
+
+    # -- list/watch (informer substrate) -------------------------------------
+    def list_raw(self, path: str, params: Optional[dict] = None) -> dict:
+        with self.lock:
+            if path.endswith("/nodes"):
+                items = [copy.deepcopy(n) for n in self.nodes.values()]
+            else:
+                items = [copy.deepcopy(p) for p in self.pods.values()]
+        return {"metadata": {"resourceVersion": str(self._rv)},
+                "items": items}
+
+    def push_watch_event(self, ev_type: str, obj: dict) -> None:
+        """Test hook: enqueue an event for watchers (and apply it to the
+        store so list/watch stay consistent)."""
+        with self.lock:
+            self._rv += 1
+            obj = copy.deepcopy(obj)
+            obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
+            meta = obj["metadata"]
+            if obj.get("kind") == "Node" or "nodeInfo" in obj.get("status", {}):
+                store, key = self.nodes, meta.get("name", "")
+            else:
+                store, key = self.pods, (meta.get("namespace", "default"),
+                                         meta.get("name", ""))
+            if ev_type == "DELETED":
+                store.pop(key, None)
+            else:
+                store[key] = obj
+            self._watch_queue.append((ev_type, obj))
+
+    def watch(self, path: str, resource_version: str,
+              params: Optional[dict] = None, timeout_s: int = 300):
+        import time as _time
+
+        deadline = _time.monotonic() + min(timeout_s, 1.0)
+        while _time.monotonic() < deadline:
+            with self.lock:
+                if self._watch_queue:
+                    yield self._watch_queue.pop(0)
+                    continue
+            _time.sleep(0.01)
 
 
 def build_kube_client() -> KubeClient:

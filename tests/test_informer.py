@@ -122,3 +122,26 @@ def test_scheduler_reads_informer_cache():
     )
     assert sched.schedule_once() == 1
     assert not d.has_topology_gate(kube.get_pod("default", "j1-0"))
+
+
+def test_informer_over_fake_kube_client():
+    """Informer end-to-end over FakeKubeClient's list/watch surface."""
+    kube = FakeKubeClient(pods=[pod("seed")])
+    inf = Informer(kube, "/api/v1/pods")
+    inf.start()
+    try:
+        assert inf.wait_synced(5)
+        assert [p["metadata"]["name"] for p in inf.items()] == ["seed"]
+        kube.push_watch_event("ADDED", pod("late"))
+        deadline = time.time() + 5
+        while len(inf) < 2 and time.time() < deadline:
+            time.sleep(0.02)
+        assert sorted(p["metadata"]["name"] for p in inf.items()) == \
+            ["late", "seed"]
+        kube.push_watch_event("DELETED", pod("seed"))
+        deadline = time.time() + 5
+        while len(inf) > 1 and time.time() < deadline:
+            time.sleep(0.02)
+        assert [p["metadata"]["name"] for p in inf.items()] == ["late"]
+    finally:
+        inf.stop()
