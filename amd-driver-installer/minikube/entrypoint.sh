@@ -108,7 +108,7 @@ install_rocm_userspace() {
 
 verify_installation() {
   # Parity: ubuntu/entrypoint.sh verify step (nvidia-smi analog).
-  "${INSTALL_DIR}/bin/rocminfo" | grep -q "gfx950"
+  "${INSTALL_DIR}/bin/rocminfo" | grep "gfx950" > /dev/null
   [[ -e /dev/kfd ]]
   compgen -G "/dev/dri/renderD*" > /dev/null
 }

@@ -110,7 +110,9 @@ verify_installation() {
   # Parity: nvidia-smi verify + device-node check, entrypoint.sh:149-156.
   [[ -e "${ROOT_MOUNT_DIR}/dev/kfd" ]] || { echo "/dev/kfd missing"; return 1; }
   ls "${ROOT_MOUNT_DIR}"/dev/dri/renderD* >/dev/null
-  "${INSTALL_DIR}/bin/rocminfo" | grep -q "gfx950"
+  # plain grep (not -q): -q exits at first match and SIGPIPEs rocminfo,
+  # which pipefail turns into rc 141
+  "${INSTALL_DIR}/bin/rocminfo" | grep "gfx950" > /dev/null
   "${INSTALL_DIR}/bin/amd-smi" list
 }
 
