@@ -89,10 +89,22 @@ def main(argv=None):
                                          f"umc={act.umc_percent:.0f}%")
         healthy &= check("driver", OK, smi.driver_version() or "unknown")
 
-        from cea_amd.deviceplugin.version_visibility import rccl_version
+        from cea_amd.deviceplugin.version_visibility import (
+            partition_capabilities,
+            rccl_version,
+        )
 
         rccl = rccl_version()
         healthy &= check("librccl", OK if rccl else WARN, rccl or "not found")
+
+        caps = partition_capabilities()
+        healthy &= check(
+            "partition modes",
+            OK if caps else WARN,
+            (f"{caps.get('amd.com/gpu.partition-modes', '?')} "
+             f"(current {caps.get('amd.com/gpu.compute-partition', '?')})"
+             if caps else "sysfs knobs not visible"),
+        )
 
         if args.drain_events_ms > 0:
             evs = smi.wait_events(args.drain_events_ms)
