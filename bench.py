@@ -216,10 +216,17 @@ def main():
     # `iters_per_graph` iterations per graph) — one replay enqueues the
     # whole captured loop, removing per-launch CPU overhead, which
     # dominates the small end of the sweep (a 1 MiB step is ~13 us eager
-    # from python, most of it launch).  RCCL collectives are capturable
-    # (comm bound via device_id above); falls back to eager if capture
-    # fails.
-    use_graphs = has_gpu and os.environ.get("CEA_BENCH_GRAPH", "1") == "1"
+    # from python, most of it launch).  Default: n==1 only.  At n>=2 the
+    # eager loop already pipelines (launch ~13 us/iter < comm time, no
+    # per-iter sync), and a capture hang across ranks would kill the one
+    # driver-run scaling measurement — robustness wins over the last few
+    # percent at the 1 MiB sweep point.  CEA_BENCH_GRAPH=1 forces capture
+    # on at any world size; =0 forces off.
+    graph_env = os.environ.get("CEA_BENCH_GRAPH", "")
+    if graph_env:
+        use_graphs = has_gpu and graph_env == "1"
+    else:
+        use_graphs = has_gpu and n == 1
 
     def make_step(buf_send, buf_recv, iters_per_graph=1):
         """Returns (callable, is_graph); the callable executes
