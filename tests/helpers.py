@@ -6,6 +6,7 @@ parity with the reference's KubeletStub (beta_plugin_test.go:37-71).
 from __future__ import annotations
 
 import os
+import time
 import threading
 from concurrent import futures
 
@@ -47,6 +48,17 @@ class KubeletStub:
     def stop(self):
         if self.server:
             self.server.stop(grace=0)
+
+    def restart_kubelet_socket(self):
+        """Simulate a kubelet restart: tear the registration server down,
+        remove the socket, and bring a fresh server up — the plugin's
+        fsnotify watcher must see the recreation and re-register (parity:
+        manager.go:534-539)."""
+        self.stop()
+        if os.path.exists(self.socket_path):
+            os.unlink(self.socket_path)
+        time.sleep(0.3)
+        self.start()
 
 
 class PluginClient:
