@@ -94,6 +94,24 @@ __global__ void copy_exact_nt(const vf4* __restrict__ src,
     __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
 }
 
+// Grid-stride nontemporal for giant buffers: the AMD dispatch packet's
+// global work size per dimension is 32-bit, so an exact-cover launch is
+// limited to 2^32-1 work items (= just under 64 GiB of float4 lanes).
+// Past that the runtime either rejects the launch (exactly 2^32: invalid
+// configuration) or — worse — the 32-bit size check wraps and the kernel
+// SILENTLY covers only n4 mod 2^32 elements (observed at 120 GiB on
+// ROCm 7.0; caught by tests/test_gpu_integration.py::
+// test_large_vram_copy_288gb_sizing).  Same measured bandwidth as the
+// exact cover (stride_nt 6564 vs exact_nt 6564 GB/s r+w,
+// profiles/copy_tune_r01.csv).
+__global__ void copy_stride_nt(const vf4* __restrict__ src,
+                               vf4* __restrict__ dst, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride)
+    __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
+}
+
 // Wave64 shuffle reduction -> per-block LDS reduction -> one atomic/block.
 __global__ void reduce_sum_f32(const float4* __restrict__ in, long n4,
                                const float* __restrict__ tail, long ntail,
@@ -203,8 +221,14 @@ void launch_copy(const void* src, void* dst, long n4, hipStream_t s) {
   } else {
     const int block = 256;
     long g = (n4 + block - 1) / block;
-    hipLaunchKernelGGL(copy_exact_nt, dim3((unsigned)g), dim3(block), 0, s,
-                       (const vf4*)src, (vf4*)dst, n4);
+    if (g * (long)block > 0xFFFFFFFFL) {
+      // 32-bit dispatch global-size limit: bounded grid-stride cover
+      hipLaunchKernelGGL(copy_stride_nt, dim3(32768), dim3(block), 0, s,
+                         (const vf4*)src, (vf4*)dst, n4);
+    } else {
+      hipLaunchKernelGGL(copy_exact_nt, dim3((unsigned)g), dim3(block), 0, s,
+                         (const vf4*)src, (vf4*)dst, n4);
+    }
   }
 }
 
