@@ -76,3 +76,29 @@ def test_allocate_p50_microbench():
 
     p50 = bench.allocate_p50_us(iters=50)
     assert 0 < p50 < 100000  # a unix-socket RPC should be well under 100 ms
+
+
+def test_bench_torchrun_launch_parity():
+    """The driver launches bench via `python -m torch.distributed.run
+    --nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 ...` — exercise
+    that exact mechanism (env propagation, rendezvous, rank-0-only JSON)
+    at world 4 on gloo."""
+    port = free_port()
+    env = dict(os.environ, PYTHONPATH=REPO)
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run",
+         "--nnodes=1", "--nproc-per-node", "4",
+         "--master-addr", "127.0.0.1", "--master-port", str(port),
+         os.path.join(REPO, "bench.py"),
+         "--gpus", "4", "--steps", "2", "--warmup", "1",
+         "--max-bytes", str(1 << 20), "--no-sweep", "--backend", "gloo"],
+        capture_output=True, text=True, timeout=300, env=env,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [l for l in proc.stdout.splitlines() if l.startswith('{"metric"')]
+    assert len(lines) == 1, "exactly one JSON line (rank 0)"
+    result = json.loads(lines[0])
+    assert result["n_gpus"] == 4
+    assert result["config"]["parallelism"] == "dp4"
+    # busbw factor for n=4 all_reduce: 2*(4-1)/4 = 1.5
+    assert abs(result["config"]["busbw_factor"] - 1.5) < 1e-6
