@@ -66,9 +66,9 @@ def distance(a: TopoKey, b: TopoKey) -> int:
     reference's bare per-level comparison (schedule-daemon.py:153-172)
     whenever label values are unique across the fleet — the real-world
     case — and, unlike bare comparison, it is a true tree metric, which
-    is what makes contiguous-window placement provably optimal
-    (hypothesis found the counterexample for bare labels: a reused host
-    label under a different subblock breaks sorted-order monotonicity)."""
+    is what lets the placement objective decompose over the topology
+    tree (assign_pods' exact DP; with bare labels a reused host label
+    under a different subblock couples distant tree branches)."""
     if a.block != b.block:
         return W_BLOCK + W_SUBBLOCK + W_HOST
     if a.subblock != b.subblock:
