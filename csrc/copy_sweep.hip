@@ -89,11 +89,12 @@ int main(int argc, char** argv) {
       {"exact_nt", 2, copy_nt_uN<2>, false},
       {"exact_nt", 4, copy_nt_uN<4>, false},
       {"exact_nt", 8, copy_nt_uN<8>, false},
+      {"gs_nt", 1, copy_nt_gs<1>, true},
       {"gs_nt", 2, copy_nt_gs<2>, true},
       {"gs_nt", 4, copy_nt_gs<4>, true},
   };
   int blocks[] = {256, 512, 1024};
-  int gs_grids[] = {8192, 16384, 32768};
+  int gs_grids[] = {16384, 32768, 65536};
 
   hipEvent_t t0, t1;
   CK(hipEventCreate(&t0));
