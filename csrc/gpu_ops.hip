@@ -223,10 +223,10 @@ void launch_copy(const void* src, void* dst, long n4, hipStream_t s) {
     long g = (n4 + block - 1) / block;
     if (g * (long)block > 0xFFFFFFFFL) {
       // 32-bit dispatch global-size limit: bounded grid-stride cover.
-      // 16384x512 is the stride_nt sweep winner (profiles/
-      // copy_tune_r01.csv: 6564 GB/s r+w, vs 6258 measured for 32768x256
-      // at 120 GiB — profiles/large_copy_bw_r02.txt).
-      hipLaunchKernelGGL(copy_stride_nt, dim3(16384), dim3(512), 0, s,
+      // Config-insensitive at these sizes: every grid/block/unroll combo
+      // lands at ~6.28 TB/s r+w @65 GiB (profiles/stride_tune_65g.csv —
+      // TLB/page-walk bound, ~5% below the 256 MiB streaming rate).
+      hipLaunchKernelGGL(copy_stride_nt, dim3(32768), dim3(256), 0, s,
                          (const vf4*)src, (vf4*)dst, n4);
     } else {
       hipLaunchKernelGGL(copy_exact_nt, dim3((unsigned)g), dim3(block), 0, s,
