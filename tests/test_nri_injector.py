@@ -226,3 +226,40 @@ def test_ttrpc_rejects_oversized_and_malformed_frames():
     assert not ep._closed.is_set()  # read loop still alive
     ep.close()
     b.close()
+
+
+def test_ttrpc_survives_garbage_frames():
+    """Adversarial wire input: random headers/bodies, oversized frame
+    lengths, and truncated streams must never hang the read loop or crash
+    the endpoint — it either ignores the junk or closes the link cleanly
+    (the containerd socket is a trust boundary for the injector)."""
+    import random
+    import socket
+    import struct
+    import time
+
+    rng = random.Random(1234)
+    for trial in range(30):
+        a, b = socket.socketpair()
+        ep = TtrpcEndpoint(a)
+        ep.register("svc", "m", lambda payload: b"")
+        ep.start()
+        try:
+            kind = trial % 3
+            if kind == 0:          # pure garbage bytes
+                b.sendall(rng.randbytes(rng.randint(1, 256)))
+            elif kind == 1:        # valid header, oversized length
+                b.sendall(HEADER.pack(1 << 30, 1, MESSAGE_TYPE_REQUEST, 0))
+            else:                  # valid header, malformed protobuf body
+                body = rng.randbytes(rng.randint(1, 64))
+                b.sendall(HEADER.pack(len(body), 3,
+                                      MESSAGE_TYPE_REQUEST, 0) + body)
+            b.shutdown(socket.SHUT_WR)
+            deadline = time.time() + 5
+            while not ep._closed.is_set() and time.time() < deadline:
+                time.sleep(0.01)
+            assert ep._closed.is_set() or kind == 2, \
+                f"trial {trial}: read loop did not terminate"
+        finally:
+            ep.close()
+            b.close()
