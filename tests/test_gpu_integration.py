@@ -481,3 +481,41 @@ def test_large_vram_copy_288gb_sizing():
     assert gbps_rw > 4000, f"{gbps_rw:.0f} GB/s r+w at 120 GiB"
     del a, b
     torch.cuda.empty_cache()
+
+
+def test_partition_set_path_noop_on_hardware():
+    """The partitioner's WRITE helper executes against real amd-smi: a
+    set to the currently-active mode succeeds through the busy-retry
+    wrapper (the same plumbing a real flip uses — command construction,
+    retry classification, success parse).  Flips to OTHER modes are
+    denied by this pool (profiles/pool_probe_r02.log) but `set SPX`
+    (profile 0) is accepted, so the write path is hardware-proven."""
+    from cea_amd.partition import partition_gpu as pg
+
+    states = pg.current_partition_status(pg.default_runner)
+    assert states, "amd-smi reported no GPUs"
+    current = states[0]["accelerator_partition"].upper()
+    pg._set_with_busy_retry(
+        pg.default_runner,
+        ["amd-smi", "set", "--gpu", "0", "--compute-partition", current])
+    after = pg.current_partition_status(pg.default_runner)
+    assert after[0]["accelerator_partition"].upper() == current
+
+
+def test_partition_run_idempotent_on_hardware(tmp_path):
+    """Full partitioner run() on real hardware with the node's CURRENT
+    mode as the desired config: parse -> status -> already-desired ->
+    no set attempted, returns False (the idempotency contract,
+    partition_gpu.go:214-220 parity, live)."""
+    import json as _json
+
+    from cea_amd.partition import partition_gpu as pg
+
+    states = pg.current_partition_status(pg.default_runner)
+    current = states[0]["accelerator_partition"].upper()
+    mem = (states[0].get("memory_partition") or "NPS1").upper()
+    cfg = tmp_path / "gpu_config.json"
+    cfg.write_text(_json.dumps(
+        {"ComputePartition": f"{current.lower()}-{mem.lower()}"}))
+    changed = pg.run(config_path=str(cfg), runner=pg.default_runner)
+    assert changed is False
