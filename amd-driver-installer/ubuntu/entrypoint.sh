@@ -17,6 +17,13 @@ set -x
 
 ROCM_VERSION="${ROCM_VERSION:-7.2}"
 AMDGPU_DRIVER_VERSION="${AMDGPU_DRIVER_VERSION:-30.20}"
+# COS DaemonSets use the single knob INSTALL_PRELOADED_ONLY=true: the KMD
+# is baked into the node image and COS has no apt — alias it onto the two
+# fine-grained flags (the image's /opt/rocm is the userspace source).
+if [[ "${INSTALL_PRELOADED_ONLY:-false}" == "true" ]]; then
+  SKIP_KMD_BUILD=1
+  SKIP_PACKAGE_INSTALL=1
+fi
 # install prefix of the ROCm userspace to stage from (versioned prefixes
 # like /opt/rocm-7.2.0 exist on some images)
 ROCM_DIR="${ROCM_DIR:-/opt/rocm}"

@@ -193,3 +193,68 @@ def test_rdma_installer_init_script_executes(tmp_path):
     assert any(l.startswith("librccl.so") for l in libs), libs
     # soname chain intact (pods dlopen librccl.so.1)
     assert any(l == "librccl.so.1" for l in libs), libs
+
+
+def test_install_preloaded_only_alias(tmp_path, mini_rocm):
+    """COS DaemonSets set the single knob INSTALL_PRELOADED_ONLY=true;
+    it must behave exactly like SKIP_KMD_BUILD+SKIP_PACKAGE_INSTALL
+    (round-2 bug: the env was set by manifests but read by nothing)."""
+    make_fake_host(tmp_path)
+    stub = "verify_installation() { :; }\nmain"
+    e = dict(os.environ)
+    e.update({
+        "ROOT_MOUNT_DIR": str(tmp_path),
+        "ROCM_DIR": str(mini_rocm),
+        "INSTALL_PRELOADED_ONLY": "true",
+        # deliberately NOT setting SKIP_* — the alias must imply them
+    })
+    script = f". '{ENTRYPOINT}'\n{stub}\n"
+    proc = subprocess.run(["bash", "-c", script], env=e,
+                          capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "install complete" in proc.stdout
+    install = tmp_path / "home" / "kubernetes" / "bin" / "amd"
+    assert any(l.startswith("librccl.so")
+               for l in os.listdir(install / "lib64"))
+    # and no apt/curl was attempted (no network in this container anyway,
+    # but the trace must not even try)
+    assert "apt-get" not in proc.stderr
+
+
+def test_manifest_env_vars_are_read_by_installer_scripts():
+    """Every env var an installer DaemonSet sets must be READ by some
+    installer script — a set-but-never-read knob silently does nothing
+    (this caught INSTALL_PRELOADED_ONLY in round 2, the same failure
+    class as the RCCL ConfigMap knob check in test_rccl_env_smoke)."""
+    import glob
+
+    import yaml
+
+    scripts = ""
+    for p in glob.glob(os.path.join(REPO, "amd-driver-installer", "**",
+                                    "entrypoint.sh"), recursive=True):
+        scripts += open(p).read()
+    manifests = glob.glob(os.path.join(REPO, "amd-driver-installer", "**",
+                                       "*.yaml"), recursive=True)
+    manifests.append(os.path.join(REPO, "daemonset.yaml"))
+    checked = 0
+    for path in manifests:
+        with open(path) as f:
+            try:
+                docs = [d for d in yaml.safe_load_all(f) if d]
+            except yaml.YAMLError:
+                continue
+        for doc in docs:
+            spec = (doc.get("spec", {}).get("template", {})
+                    .get("spec", {})) if doc.get("kind") == "DaemonSet" else {}
+            for c in (spec.get("initContainers", [])
+                      + spec.get("containers", [])):
+                if "driver-installer" not in c.get("image", ""):
+                    continue
+                for env in c.get("env", []) or []:
+                    name = env.get("name", "")
+                    assert name in scripts, (
+                        f"{os.path.relpath(path, REPO)} sets {name} but no "
+                        "installer entrypoint reads it")
+                    checked += 1
+    assert checked >= 3  # the check actually saw env vars
