@@ -137,36 +137,36 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
 
     def combine(children, bonus_w):
         """Knapsack-merge child (vals, rec) pairs; add bonus_w*C(k,2).
-        Returns (vals over 0..G, rec(k) -> {hostpath: count})."""
-        acc = [0] + [NEG] * G
+        Returns (vals over 0..G, rec(k) -> {hostpath: count}).  The
+        (max,+) convolution per child is vectorized with numpy — one
+        shifted row per child count j — keeping the 512-pod/2048-node
+        case well under a second."""
+        import numpy as np
+
+        acc = np.full(G + 1, NEG)
+        acc[0] = 0.0
         acc_cap = 0
         backs = []
         for vals, _rec in children:
             child_cap = len(vals) - 1
             new_cap = min(acc_cap + child_cap, G)
-            new = [NEG] * (G + 1)
-            back = [0] * (G + 1)
-            for k in range(new_cap + 1):
-                best = NEG
-                bj = 0
-                jlo = max(0, k - acc_cap)
-                for j in range(jlo, min(k, child_cap) + 1):
-                    a = acc[k - j]
-                    if a == NEG:
-                        continue
-                    v = a + vals[j]
-                    if v > best:
-                        best, bj = v, j
-                new[k] = best
-                back[k] = bj
+            cand = np.full((child_cap + 1, G + 1), NEG)
+            for j in range(child_cap + 1):
+                lo = j
+                src_hi = min(acc_cap, new_cap - j) + 1
+                if src_hi <= 0:
+                    continue
+                cand[j, lo:lo + src_hi] = acc[0:src_hi] + vals[j]
+            best = cand.max(axis=0)
+            back = cand.argmax(axis=0)
             backs.append(back)
-            acc = new
+            acc = best
             acc_cap = new_cap
 
-        out_vals = [
-            acc[k] + bonus_w * pairs(k) if acc[k] != NEG else NEG
-            for k in range(G + 1)
-        ]
+        bonus = np.array([bonus_w * pairs(k) for k in range(G + 1)],
+                         dtype=float)
+        out_np = np.where(np.isfinite(acc), acc + bonus, NEG)
+        out_vals = out_np.tolist()
 
         def rec(k: int) -> Dict[Tuple[str, str, str], int]:
             out: Dict[Tuple[str, str, str], int] = {}
@@ -180,11 +180,37 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
 
         return out_vals, rec
 
-    def leaf(path, cap):
-        vals = [W_HOST * pairs(k) for k in range(cap + 1)]
+    def hosts_greedy(host_caps):
+        """All hosts of one subblock share the weight W_HOST, so the
+        max of sum(W_HOST*C(k_h,2)) subject to sum k_h = k, k_h <= cap
+        concentrates: fill hosts largest-capacity-first (each next unit
+        goes where the marginal C(f+1,2)-C(f,2) = f is largest, i.e. the
+        host already being filled).  O(H log H + G) replacing the
+        host-level knapsack — the dominant cost at cluster scale."""
+        caps = sorted(host_caps.items(), key=lambda kv: (-kv[1], kv[0]))
+        K = min(sum(c for _, c in caps), G)
+        vals = [0] * (K + 1)
+        bonus = 0
+        hi = 0
+        fill = 0
+        for k in range(1, K + 1):
+            while hi < len(caps) and fill >= caps[hi][1]:
+                hi += 1
+                fill = 0
+            bonus += W_HOST * fill   # marginal of C(fill+1,2)
+            fill += 1
+            vals[k] = bonus
 
         def rec(k: int) -> Dict[Tuple[str, str, str], int]:
-            return {path: k} if k else {}
+            out: Dict[Tuple[str, str, str], int] = {}
+            for path, cap in caps:
+                take = min(cap, k)
+                if take:
+                    out[path] = take
+                k -= take
+                if k == 0:
+                    break
+            return out
 
         return vals, rec
 
@@ -192,9 +218,12 @@ def assign_pods(num_pods: int, nodes: List[CandidateNode]
     for b in sorted(tree):
         sub_entries = []
         for s in sorted(tree[b]):
-            hosts = [leaf(path, cap)
-                     for path, cap in sorted(tree[b][s].items())]
-            sub_entries.append(combine(hosts, W_SUBBLOCK))
+            hosts_vals, hosts_rec = hosts_greedy(tree[b][s])
+            sub_vals = [
+                v + W_SUBBLOCK * pairs(k) if v != NEG else NEG
+                for k, v in enumerate(hosts_vals)
+            ]
+            sub_entries.append((sub_vals, hosts_rec))
         block_entries.append(combine(sub_entries, W_BLOCK))
     root_vals, root_rec = combine(block_entries, 0)
     if root_vals[G] == NEG:
