@@ -235,7 +235,6 @@ def test_ttrpc_survives_garbage_frames():
     (the containerd socket is a trust boundary for the injector)."""
     import random
     import socket
-    import struct
     import time
 
     rng = random.Random(1234)
