@@ -519,3 +519,43 @@ def test_partition_run_idempotent_on_hardware(tmp_path):
         {"ComputePartition": f"{current.lower()}-{mem.lower()}"}))
     changed = pg.run(config_path=str(cfg), runner=pg.default_runner)
     assert changed is False
+
+
+def test_windowed_utilization_tracks_duty_cycle():
+    """Quantitative sampler check: a ~50% duty load (alternating busy and
+    idle phases) must land the 16s-style windowed average in a band
+    around 50% — validating that the C++ sampler's average reflects the
+    actual duty cycle, not just load presence (the reference's
+    nvmlDeviceGetAverageUsage contract, metrics/util.go:34-88)."""
+    import threading
+
+    import cea_amd.amdsmi as amdsmi
+    from cea_amd.amdsmi.shim import ShimAmdSmi
+
+    smi = ShimAmdSmi(sampler_interval_ms=100)
+    smi.init()
+    try:
+        stop = threading.Event()
+
+        def duty_burn():
+            a = torch.randn(4096, 4096, device="cuda")
+            torch.cuda.synchronize()
+            while not stop.is_set():
+                t_end = time.time() + 0.5
+                while time.time() < t_end:      # busy half-period
+                    a = a @ a
+                    a = a / a.norm()
+                torch.cuda.synchronize()
+                time.sleep(0.5)                  # idle half-period
+
+        t = threading.Thread(target=duty_burn, daemon=True)
+        t.start()
+        time.sleep(10)  # ~100 samples across ~10 duty periods
+        busy = smi.average_gfx_utilization(0, 8.0)
+        stop.set()
+        t.join(timeout=30)
+        # generous band: phase boundaries + amdsmi's own sampling lag
+        assert 25 <= busy <= 80, f"windowed avg {busy}% for a ~50% duty load"
+    finally:
+        smi.shutdown()
+        amdsmi.ops = None
