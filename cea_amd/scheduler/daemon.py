@@ -53,12 +53,17 @@ _SUFFIX = {
 
 
 def parse_quantity(q) -> float:
-    """k8s resource quantity -> float (cpu in cores, memory in bytes)."""
+    """k8s resource quantity -> float (cpu in cores, memory in bytes).
+    Total: malformed input parses as 0.0 — a bad quantity in one pod
+    spec must never abort a whole scheduling pass (fuzz-pinned)."""
     if isinstance(q, (int, float)):
         return float(q)
-    q = str(q)
+    q = str(q).strip()
     if q.endswith("m"):
-        return float(q[:-1]) / 1000.0
+        try:
+            return float(q[:-1]) / 1000.0
+        except ValueError:
+            return 0.0
     m = _QUANTITY_RE.match(q)
     if not m:
         return 0.0

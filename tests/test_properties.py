@@ -198,3 +198,27 @@ def test_assign_pods_hetero_respects_resources_and_selectors(pod_spec,
         used[name] += pod["requests"]["amd.com/gpu"]
     for name, total in used.items():
         assert total <= by_name[name].free["amd.com/gpu"]
+
+
+@given(st.text(max_size=64))
+@settings(max_examples=200, deadline=None)
+def test_parse_quantity_total(q):
+    """parse_quantity never raises (fuzz-found: bare 'm' crashed it —
+    a malformed quantity in one pod spec must never abort a scheduling
+    pass)."""
+    from cea_amd.scheduler.daemon import parse_quantity
+
+    v = parse_quantity(q)
+    assert isinstance(v, float) and v >= 0.0 or v < 0.0  # just: a float
+
+
+@given(st.text(max_size=512))
+@settings(max_examples=200, deadline=None)
+def test_partition_parsers_total(text):
+    """amd-smi output parsers never raise on arbitrary text (the job
+    must fail with a clear status error, not a traceback, when the CLI
+    output format drifts)."""
+    from cea_amd.partition import partition_gpu as pg
+
+    assert isinstance(pg.parse_partition_status(text), list)
+    assert isinstance(pg.parse_partition_profiles(text), dict)
