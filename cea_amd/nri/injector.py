@@ -65,6 +65,10 @@ def get_devices(ctr_name: str, pod_annotations: Dict[str, str]) -> List[dict]:
         if not isinstance(entry, dict) or not entry.get("path"):
             raise DeviceError(f"device entry missing path: {entry!r}")
         path = entry["path"]
+        if not isinstance(path, str):
+            # YAML can produce ints/lists/dicts here; anything non-string
+            # is a malformed annotation, not a device path
+            raise DeviceError(f"device path must be a string: {path!r}")
         if path in seen:
             continue
         seen.add(path)

@@ -262,3 +262,14 @@ def test_ttrpc_survives_garbage_frames():
         finally:
             ep.close()
             b.close()
+
+
+def test_get_devices_rejects_non_string_paths():
+    """Malformed annotations with non-string path values (ints, lists,
+    dicts from YAML) raise DeviceError — not TypeError (fuzz-found:
+    unhashable paths crashed the dedup set)."""
+    from cea_amd.nri.injector import DeviceError, get_devices
+
+    for val in ("- path: [a, b]", "- path: {x: 1}", "- path: 3"):
+        with pytest.raises(DeviceError):
+            get_devices("c", {"devices.gke.io/container.c": val})
