@@ -218,3 +218,41 @@ def test_no_dead_env_knobs_across_all_manifests():
                         "cea-amd image but nothing reads it")
                     checked += 1
     assert checked >= 6, checked
+
+
+def test_event_config_codes_are_known():
+    """Every code in the shipped event ConfigMaps must be a code the
+    health stack can actually raise (EVT_* constants) — a nonsense code
+    silently never fires (round-2: the example shipped 5, the benign
+    MIGRATE_START that is deliberately not even armed)."""
+    import glob
+
+    import yaml
+
+    from cea_amd.amdsmi import iface
+
+    known = {
+        v for k, v in vars(iface).items() if k.startswith("EVT_")
+    }
+    found = 0
+    for path in glob.glob(os.path.join(REPO, "deploy", "**", "*.yaml"),
+                          recursive=True) + glob.glob(
+                              os.path.join(REPO, "test", "**", "*.yaml"),
+                              recursive=True):
+        with open(path) as f:
+            try:
+                docs = [d for d in yaml.safe_load_all(f) if d]
+            except yaml.YAMLError:
+                continue
+        for doc in docs:
+            if doc.get("kind") != "ConfigMap":
+                continue
+            val = (doc.get("data") or {}).get("health-critical-events")
+            if not val:
+                continue
+            found += 1
+            for code in str(val).split(","):
+                assert int(code) in known, (
+                    f"{os.path.relpath(path, REPO)}: event code {code} is "
+                    f"not a known EVT_* code {sorted(known)}")
+    assert found >= 1
