@@ -256,3 +256,37 @@ def test_event_config_codes_are_known():
                     f"{os.path.relpath(path, REPO)}: event code {code} is "
                     f"not a known EVT_* code {sorted(known)}")
     assert found >= 1
+
+
+def test_gpu_config_examples_parse():
+    """Every shipped gpu_config example must parse through the real
+    config parser (round-2: dpx/qpx examples shipped an NPS pairing the
+    hardware profile table does not validate)."""
+    import json
+
+    import yaml
+
+    from cea_amd.partition.partition_gpu import (
+        PARTITION_COUNT,
+        parse_partition_config,
+    )
+
+    path = os.path.join(REPO, "deploy", "device-plugin",
+                        "gpu-config-examples.yaml")
+    doc = next(d for d in yaml.safe_load_all(open(path)) if d)
+    import tempfile
+    n = 0
+    for key, raw in doc["data"].items():
+        json.loads(raw)  # valid JSON
+        with tempfile.NamedTemporaryFile("w", suffix=".json",
+                                         delete=False) as f:
+            f.write(raw)
+            p = f.name
+        compute, memory = parse_partition_config(p)
+        assert compute in PARTITION_COUNT, (key, compute)
+        # NPS1 is the hardware-validated pairing for every compute mode
+        # (pool_probe_r02); examples must not ship unvalidated pairings
+        if compute != "SPX":
+            assert memory == "NPS1", (key, memory)
+        n += 1
+    assert n >= 5
