@@ -582,3 +582,72 @@ def test_health_detection_latency_bound():
         v = hl.one_run(env, tmp)
     assert v is not None, "Unhealthy resend never arrived"
     assert v < 5.0, f"detection took {v:.2f}s (> event-wait bound)"
+
+
+def _inline_python(yaml_path, kind_name):
+    import yaml as _yaml
+
+    with open(yaml_path) as f:
+        docs = [d for d in _yaml.safe_load_all(f) if d]
+    for doc in docs:
+        if doc.get("metadata", {}).get("name") != kind_name:
+            continue
+        spec = doc["spec"]
+        pod = spec.get("template", {}).get("spec", spec)
+        for c in pod.get("containers", []):
+            cmd = c.get("command", [])
+            if cmd[:2] == ["python3", "-c"]:
+                return cmd[2]
+    raise AssertionError(f"{kind_name} python3 -c block not found")
+
+
+def test_training_demo_command_runs_on_hardware(tmp_path):
+    """The training demo's in-container command (demo/gpu-training/
+    train-synthetic.yaml) executes verbatim on a real MI355X — the demo
+    is runnable, not decorative (parity: the reference's demo training
+    jobs are known-good images)."""
+    code = _inline_python(
+        os.path.join(REPO, "demo", "gpu-training", "train-synthetic.yaml"),
+        "amd-gpu-train-demo")
+    proc = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                          text=True, timeout=600)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "demo training done" in proc.stdout
+    # loss decreased over the run
+    import re
+    losses = [float(m) for m in re.findall(r"loss ([0-9.]+)", proc.stdout)]
+    assert len(losses) >= 2 and losses[-1] < losses[0], losses
+
+
+def test_serving_demo_command_runs_on_hardware(tmp_path):
+    """The serving demo's FastAPI server comes up on hardware and answers
+    a /predict round trip."""
+    import json as _json
+    import urllib.request
+
+    code = _inline_python(
+        os.path.join(REPO, "demo", "gpu-serving", "serving.yaml"),
+        "amd-gpu-serving-demo")
+    code = code.replace("port=8500", "port=18500")
+    proc = subprocess.Popen([sys.executable, "-c", code],
+                            stdout=subprocess.PIPE,
+                            stderr=subprocess.STDOUT, text=True)
+    try:
+        deadline = time.time() + 120
+        body = None
+        while time.time() < deadline:
+            try:
+                req = urllib.request.Request(
+                    "http://127.0.0.1:18500/predict?n=4", method="POST")
+                body = _json.loads(
+                    urllib.request.urlopen(req, timeout=5).read())
+                break
+            except Exception:
+                time.sleep(1)
+        assert body == {"shape": [4, 1024]}, body
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
